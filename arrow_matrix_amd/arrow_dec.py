@@ -1,0 +1,290 @@
+"""ArrowDecompositionMPI — multi-matrix orchestration of the hot path.
+
+Re-implements the semantics of the reference's `arrow/arrow_dec_mpi.py`
+(class of the same name): per iteration
+
+  1. forward feature propagation matrix i -> i+1 along the composed
+     permutations (_propagate_features_forwards, arrow_dec_mpi.py:507-610),
+  2. per-matrix slim arrow SpMM (arrow_slim.py),
+  3. backward partial-result aggregation matrix i -> i-1 with scatter-add
+     (_aggregate_features_backwards, arrow_dec_mpi.py:404-505).
+
+MI355X-first deviations:
+  * SHARED-RANK layout: every rank hosts a contiguous stripe of EVERY
+    decomposition part (the reference gives each part a disjoint rank set,
+    needing sum(n_blocks) ranks). All parts' SpMMs run on all GPUs; the
+    inter-part permutation exchange becomes ONE alltoallv over the node's
+    xGMI links, with the rank-local share never leaving HBM.
+    Consequently matrix_index == 0 on every rank and `B` is the part-0
+    engine (the part that carries X, the reference's postcondition).
+  * Permutation gathers/scatters run as device kernels (C ABI) instead of
+    host fancy-indexing (arrow_dec_mpi.py:421,437,526,544).
+  * The loader reads the decomposition on every rank (one node, shared
+    filesystem, mmap) instead of root-scatter over MPI.
+"""
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from . import graphio, tables
+from .arrow_matrix import ArrowMatrix
+from .arrow_slim import ArrowSlimMPI
+from .comm import Comm, default_comm
+from .common import wb_logging
+import time
+
+
+class _Exchange:
+    """Routing tables + index tensors for one matrix pair (i -> i+1)."""
+
+    def __init__(self, send_counts, send_rows, recv_counts, recv_rows, backend):
+        self.send_counts = [int(c) for c in send_counts]
+        self.recv_counts = [int(c) for c in recv_counts]
+        self.send_rows = backend.index_tensor(send_rows)
+        self.recv_rows = backend.index_tensor(recv_rows)
+
+
+class ArrowDecompositionMPI:
+
+    def __init__(self, comm: Comm, engines: List[ArrowSlimMPI], n_blocks: np.ndarray,
+                 width: int, n_feature_columns: int,
+                 to_prev: List[Optional[np.ndarray]], to_next: List[Optional[np.ndarray]],
+                 device: str = 'cpu'):
+        self.comm = comm
+        self.engines = engines
+        self.n_blocks = np.asarray(n_blocks)
+        self.decomposition_length = len(engines)
+        self.matrix_index = 0  # shared-rank layout: every rank hosts part 0
+        self.device = device
+        self._n_rows_per_rank = width
+        self._n_feature_columns = n_feature_columns
+        self.width = width
+        self._to_prev = to_prev
+        self._to_next = to_next
+        self._forward: List[Optional[_Exchange]] = []
+        self._backward: List[Optional[_Exchange]] = []
+        self._initialize_all_to_all_tables()
+
+    @property
+    def B(self) -> ArrowMatrix:
+        return self.engines[0]
+
+    # -- setup ---------------------------------------------------------------
+
+    @staticmethod
+    def initialize(comm: Optional[Comm], n_blocks: np.ndarray,
+                   to_prev_permutation, to_next_permutation,
+                   rows_per_rank: int, feature_columns: int, device: str = 'cpu',
+                   block_diagonal: bool = True, slim: bool = True):
+        """Factory, keeping the reference's signature
+        (arrow_dec_mpi.py:106-177). to_prev/to_next are the per-rank
+        permutation slices as returned by load_decomposition_new (lists with
+        one entry per decomposition part)."""
+        assert slim and block_diagonal, \
+            "round 1 implements the slim block-diagonal path (reference default); " \
+            "ArrowMPI split/banded mode is tracked in DESIGN.md §next"
+        comm = comm if comm is not None else default_comm()
+        engines = [ArrowSlimMPI(comm, tiles_per_side=int(nb), device=device)
+                   for nb in n_blocks]
+        return ArrowDecompositionMPI(comm, engines, n_blocks, rows_per_rank,
+                                     feature_columns, to_prev_permutation,
+                                     to_next_permutation, device=device)
+
+    def _initialize_all_to_all_tables(self) -> None:
+        """Build the forward/backward exchange tables
+        (reference arrow_dec_mpi.py:210-281,325-384, generalised to the
+        shared-rank layout in tables.routing_*_tables)."""
+        P = self.comm.size
+        w = self.width
+        L = self.decomposition_length
+        self._forward = [None] * L
+        self._backward = [None] * L
+        for i in range(L - 1):
+            eng_s, eng_r = self.engines[i], self.engines[i + 1]
+            own_s = tables.contiguous_block_owners(int(self.n_blocks[i]), P)
+            own_r = tables.contiguous_block_owners(int(self.n_blocks[i + 1]), P)
+            # forward (features i -> i+1): sender tables on my to_next slice
+            # of matrix i, receiver tables on my to_prev slice of matrix i+1
+            s_cnt, s_rows = tables.routing_send_tables(
+                self._slice_for_rank(self._to_next[i], eng_s), w, own_r,
+                int(self.n_blocks[i + 1]), P)
+            r_cnt, r_rows = tables.routing_recv_tables(
+                self._slice_for_rank(self._to_prev[i + 1], eng_r), w, own_s,
+                int(self.n_blocks[i]), P)
+            self._forward[i] = _Exchange(s_cnt, s_rows, r_cnt, r_rows, eng_s.backend)
+            # backward (partials i+1 -> i): sender tables on my to_prev slice
+            # of matrix i+1, receiver tables on my to_next slice of matrix i
+            s_cnt, s_rows = tables.routing_send_tables(
+                self._slice_for_rank(self._to_prev[i + 1], eng_r), w, own_s,
+                int(self.n_blocks[i]), P)
+            r_cnt, r_rows = tables.routing_recv_tables(
+                self._slice_for_rank(self._to_next[i], eng_s), w, own_r,
+                int(self.n_blocks[i + 1]), P)
+            self._backward[i + 1] = _Exchange(s_cnt, s_rows, r_cnt, r_rows, eng_r.backend)
+
+    def _slice_for_rank(self, perm_full: np.ndarray, engine: ArrowSlimMPI) -> np.ndarray:
+        """This rank's rows of a matrix = its contiguous block span."""
+        w = self.width
+        return perm_full[engine.first_block * w: engine.last_block * w]
+
+    def load_data_from_blocks(self, blocked) -> None:
+        """blocked: one block grid per decomposition part
+        (reference arrow_dec_mpi.py:179-181)."""
+        assert len(blocked) == self.decomposition_length
+        for eng, blocks in zip(self.engines, blocked):
+            eng.load_sparse_matrix_from_blocks(blocks)
+
+    def zero_rhs(self, width: int, n_features: int) -> None:
+        for eng in self.engines:
+            eng.zero_rhs(width, n_features)
+
+    # -- iteration -----------------------------------------------------------
+
+    def step(self) -> None:
+        """One X <- A @ X iteration (reference arrow_dec_mpi.py:283-307)."""
+        tic = time.perf_counter()
+        self._propagate_features()
+        wb_logging.log({"spmm_bcast_time": time.perf_counter() - tic})
+
+        tic = time.perf_counter()
+        for eng in self.engines:
+            eng.spmm()
+        wb_logging.log({'spmm_arrow_time': time.perf_counter() - tic})
+
+        tic = time.perf_counter()
+        self._aggregate()
+        wb_logging.log({"spmm_reduce_time": time.perf_counter() - tic})
+
+    def _propagate_features(self) -> None:
+        """Forward: route matrix i's features to matrix i+1 along the
+        composed permutation (arrow_dec_mpi.py:507-610)."""
+        for i in range(self.decomposition_length - 1):
+            ex = self._forward[i]
+            eng_s, eng_r = self.engines[i], self.engines[i + 1]
+            sendbuf = eng_s.backend.gather_rows(eng_s.feature_tile(), ex.send_rows)
+            recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
+            # C_i[recv_perm] = recvbuf; X := C (arrow_dec_mpi.py:544-545)
+            eng_r.backend.scatter_rows(eng_r.C_i, ex.recv_rows, recvbuf)
+            eng_r.set_features(eng_r.C_i)
+
+    def _aggregate(self) -> None:
+        """Backward: cascade partial results matrix i -> i-1 with
+        scatter-add (arrow_dec_mpi.py:404-505)."""
+        for i in reversed(range(1, self.decomposition_length)):
+            ex = self._backward[i]
+            eng_s, eng_r = self.engines[i], self.engines[i - 1]
+            sendbuf = eng_s.backend.gather_rows(eng_s.result_tile(), ex.send_rows)
+            recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
+            # C_i[recv_perm] += recvbuf; X := C (arrow_dec_mpi.py:437-438)
+            eng_r.backend.scatter_add_rows(eng_r.C_i, ex.recv_rows, recvbuf)
+            eng_r.set_features(eng_r.C_i)
+
+    # -- loading -------------------------------------------------------------
+
+    @staticmethod
+    def number_of_blocks(adjacency, width: int) -> int:
+        """Reference arrow_dec_mpi.py:612-627."""
+        if isinstance(adjacency, tuple):
+            indptr = adjacency[2]
+            nnz_per_row = np.asarray(indptr[1:]) - np.asarray(indptr[:-1])
+        else:
+            nnz_per_row = adjacency.getnnz(1)
+        return tables.number_of_blocks(nnz_per_row, width)
+
+    @staticmethod
+    def load_decomposition_new(comm: Optional[Comm], filename: str, width: int,
+                               is_block_diagonal: bool = True, datatype=np.float32,
+                               slim: bool = True, use_npy: bool = True,
+                               use_mmap: bool = True):
+        """Load a decomposed graph and build this rank's blocks + permutation
+        slices. Keeps the reference's signature and return shape
+        (arrow_dec_mpi.py:629-930): returns
+        (blocks, n_blocks, to_prev, to_next) — here `blocks` is one grid per
+        part and to_prev/to_next are per-part lists of this rank's slices
+        (the shared-rank layout hosts every part on every rank).
+
+        Every rank reads the files itself (single node, shared FS, mmap) —
+        no root scatter. Zero-block cutting, permutation padding /
+        one-based normalisation / overflow sentinel follow
+        arrow_dec_mpi.py:680-749 exactly (tables.pad_and_compose_permutations).
+        """
+        if not use_npy:
+            raise NotImplementedError(
+                "only the .npy CSR format is supported (use_npy=True); the "
+                "legacy .npz path is tracked in DESIGN.md §next")
+        comm = comm if comm is not None else default_comm()
+        P = comm.size
+        rank = comm.rank
+
+        decomposition = graphio.load_decomposition_new(
+            filename, width, block_diagonal=is_block_diagonal, mem_map=use_mmap)
+        if len(decomposition) == 0:
+            raise FileNotFoundError(
+                f"decomposition {filename!r} width={width} not found")
+
+        n_blocks = np.zeros(len(decomposition), dtype=np.int32)
+        for i, (adjacency, _) in enumerate(decomposition):
+            n_blocks[i] = ArrowDecompositionMPI.number_of_blocks(adjacency, width)
+
+        perms = [p for _, p in decomposition]
+        _, to_prev_full, to_next_full = tables.pad_and_compose_permutations(
+            perms, n_blocks, width)
+
+        blocks_per_part = []
+        to_prev_sliced: List[Optional[np.ndarray]] = []
+        to_next_sliced: List[Optional[np.ndarray]] = []
+        for i, (adjacency, _) in enumerate(decomposition):
+            nb = int(n_blocks[i])
+            bpr = -(-nb // P)
+            first = min(rank * bpr, nb)
+            last = min(first + bpr, nb)
+            from scipy import sparse
+            if isinstance(adjacency, tuple):
+                data, indices, indptr = adjacency
+            else:
+                data, indices, indptr = adjacency.data, adjacency.indices, adjacency.indptr
+            # adjacency matrices are square (reference asserts this,
+            # graphio.py:427); build with the explicit n x n shape instead of
+            # inferring cols from max(indices)+1 as csr_matrix would
+            n_rows = indptr.size - 1
+            A = sparse.csr_matrix((data, indices, indptr), shape=(n_rows, n_rows))
+            A = A.astype(datatype)
+            grid: List[List[Optional[object]]] = [[None] * nb for _ in range(nb)]
+            # first block-row tiles for owned columns; diagonal + first
+            # block-column for owned rows (slim layout,
+            # arrow_slim_mpi.py:298-326)
+            for c in range(first, last):
+                grid[0][c] = _extract_block(A, 0, c, width)
+            for r in range(max(first, 1), last):
+                grid[r][r] = _extract_block(A, r, r, width)
+                grid[r][0] = _extract_block(A, r, 0, width)
+            blocks_per_part.append(grid)
+            w = width
+            tp = to_prev_full[i]
+            tn = to_next_full[i]
+            to_prev_sliced.append(None if tp is None else tp[first * w:last * w])
+            to_next_sliced.append(None if tn is None else tn[first * w:last * w])
+
+        return blocks_per_part, n_blocks, to_prev_sliced, to_next_sliced
+
+
+def _extract_block(A, i: int, j: int, block_size: int):
+    """Extract block (i, j) with the reference's split semantics
+    (graphio.py:361-406 / load_block_from_bslice:449-495): short last
+    block-row edge-padded to block_size and declared square."""
+    from scipy import sparse
+    rows, cols = A.shape
+    r0, r1 = i * block_size, min(rows, (i + 1) * block_size)
+    c0, c1 = j * block_size, min(cols, (j + 1) * block_size)
+    sl = A[r0:r1, c0:c1]
+    pad_width = block_size - (r1 - r0)
+    if pad_width == 0:
+        block = sparse.csr_matrix(sl)
+    else:
+        indx_ptr = np.pad(sl.indptr, (0, pad_width), mode='edge')
+        block = sparse.csr_matrix((sl.data, sl.indices, indx_ptr),
+                                  shape=(block_size, block_size))
+    block.sum_duplicates()
+    block.sort_indices()
+    return block

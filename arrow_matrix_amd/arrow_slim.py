@@ -1,0 +1,187 @@
+"""Slim arrow SpMM engine — the per-matrix compute of the hot path.
+
+Re-implements the semantics of the reference's `ArrowSlimMPI`
+(arrow_slim_mpi.py): block-row i of an arrow matrix holds A_0i (first
+block-row), A_ii (diagonal) and A_i0 (first block-column); per iteration
+
+    C_0 = sum_i A_0i @ X_i            (reduced to the owner of block-row 0,
+                                       arrow_slim_mpi.py:104-116)
+    C_i = A_ii @ X_i + A_i0 @ X_0     (arrow_slim_mpi.py:121-144)
+
+with X_0 broadcast (arrow_slim_mpi.py:269-273).
+
+MI355X-first deviations from the reference:
+  * A rank owns a CONTIGUOUS SPAN of block-rows, not exactly one: 288 GB of
+    HBM3E per GPU holds many width-wide tiles, so 1..8 ranks cover any
+    decomposition (the reference needs sum(n_blocks) MPI ranks). At one
+    block per rank this reduces to the reference layout.
+  * Partial sums of C_0 over the rank's own blocks are accumulated on-GPU
+    first; ONE reduce follows (the reference reduces one partial per rank).
+  * A is uploaded to HBM once at load time and stays resident (the reference
+    re-uploads every call, arrow_slim_mpi.py:184-232 — noted TODO at
+    arrow_mpi.py:314).
+  * Collectives are RCCL over xGMI (comm.py) instead of mpi4py.
+"""
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from .arrow_matrix import ArrowMatrix
+from .backends import make_backend
+from .comm import Comm
+
+
+class ArrowSlimMPI(ArrowMatrix):
+
+    def __init__(self, comm: Optional[Comm] = None, tiles_per_side: Optional[int] = None,
+                 device: str = 'cpu'):
+        """:param comm: communicator over this matrix's ranks (None = single
+        process). :param tiles_per_side: number of block-rows (n_blocks);
+        defaults to comm.size, the reference's one-rank-per-block layout."""
+        self.comm = comm if comm is not None else Comm()
+        self.column_comm = self.comm
+        self.tiles_per_side = tiles_per_side if tiles_per_side is not None else self.comm.size
+        assert self.tiles_per_side >= 1
+        self.device = device
+        self.backend = make_backend(device)
+
+        # contiguous block span owned by this rank
+        bpr = -(-self.tiles_per_side // self.comm.size)
+        self.blocks_per_rank = bpr
+        self.first_block = min(self.comm.rank * bpr, self.tiles_per_side)
+        self.last_block = min(self.first_block + bpr, self.tiles_per_side)
+        self.n_owned = self.last_block - self.first_block
+
+        self.width: Optional[int] = None
+        self.A_0i: List = []          # A_0c for owned c (product handles or scipy)
+        self.A_ii: List = []          # A_rr for owned r > 0 (None at r == 0)
+        self.A_i0: List = []          # A_r0 for owned r > 0
+        self.X_i: Optional[torch.Tensor] = None   # (n_owned*width, k) stripe
+        self.X_0: Optional[torch.Tensor] = None   # (width, k)
+        self.C_i: Optional[torch.Tensor] = None   # (n_owned*width, k) stripe
+        self.C_0: Optional[torch.Tensor] = None   # (width, k) partial/reduced
+        self.nnz_owned = 0
+
+    # -- data loading --------------------------------------------------------
+
+    def load_sparse_matrix_from_blocks(self, blocks) -> None:
+        """blocks: block grid (list of lists; entries not owned by this rank
+        may be None). Owned pieces are uploaded once and stay resident."""
+        assert len(blocks) == self.tiles_per_side
+        self.A_0i, self.A_ii, self.A_i0 = [], [], []
+        self.nnz_owned = 0
+        for r in range(self.first_block, self.last_block):
+            b0r = blocks[0][r]
+            assert b0r is not None, f"missing block (0,{r})"
+            if self.width is None:
+                self.width = b0r.shape[0]
+            self.A_0i.append(self.backend.upload_block(b0r))
+            self.nnz_owned += b0r.nnz
+            if r > 0:
+                assert blocks[r][r] is not None and blocks[r][0] is not None, \
+                    f"missing diagonal/column block for row {r}"
+                self.A_ii.append(self.backend.upload_block(blocks[r][r]))
+                self.A_i0.append(self.backend.upload_block(blocks[r][0]))
+                self.nnz_owned += blocks[r][r].nnz + blocks[r][0].nnz
+            else:
+                self.A_ii.append(None)
+                self.A_i0.append(None)
+
+    # -- buffers -------------------------------------------------------------
+
+    def zero_rhs(self, number_of_rows_per_rank: int, number_of_columns: int,
+                 dtype=np.float32) -> None:
+        assert number_of_rows_per_rank >= 1 and number_of_columns >= 1
+        self.width = number_of_rows_per_rank
+        w, k = number_of_rows_per_rank, number_of_columns
+        stripe = (max(self.n_owned, 1) * w, k)
+        for name, shape in (('C_i', stripe), ('C_0', (w, k)),
+                            ('X_i', stripe), ('X_0', (w, k))):
+            buf = getattr(self, name)
+            if buf is None or tuple(buf.shape) != shape:
+                setattr(self, name, self.backend.zeros(shape))
+            else:
+                buf.zero_()
+
+    def set_features(self, X) -> None:
+        """Stores a reference (no copy) when X already lives on this
+        engine's device; otherwise moves it there once."""
+        assert X is not None
+        if isinstance(X, torch.Tensor) and (X.device.type == self.backend.device
+                                            or (self.backend.device == 'cuda' and X.is_cuda)):
+            self.X_i = X
+        else:
+            self.X_i = self.backend.asarray(X)
+
+    def feature_tile(self):
+        return self.X_i
+
+    def result_tile(self):
+        return self.C_i
+
+    def is_column_rank(self) -> bool:
+        return True
+
+    # -- compute -------------------------------------------------------------
+
+    def spmm(self, device: str = None) -> None:
+        """One arrow SpMM (reference _arrow_spmm, arrow_slim_mpi.py:246-280).
+        `device` is accepted for API compatibility; the engine's device was
+        fixed at construction (no silent fallback)."""
+        if device is not None and device != self.device:
+            raise ValueError(
+                f"engine was built for device={self.device!r}; got {device!r}")
+        be = self.backend
+        w = self.width
+        k = self.X_i.shape[1]
+
+        # X_0 broadcast (arrow_slim_mpi.py:265-273); owner of block 0 is rank 0
+        if self.first_block == 0 and self.n_owned > 0:
+            self.X_0.copy_(self.X_i[:w])
+        self.comm.bcast_(self.X_0, src=0)
+
+        # C_0 partial: on-GPU accumulation over owned first-block-row tiles
+        first = True
+        for j, r in enumerate(range(self.first_block, self.last_block)):
+            Xr = self.X_i[j * w:(j + 1) * w]
+            be.spmm_block(self.A_0i[j], Xr.contiguous(), self.C_0, beta=0 if first else 1)
+            first = False
+        if first:  # rank owns no blocks of this matrix
+            self.C_0.zero_()
+
+        # ONE reduce of the first block-row partials (arrow_slim_mpi.py:116)
+        self.comm.reduce_sum_(self.C_0, dst=0)
+
+        # Diagonal + first-block-column tiles (arrow_slim_mpi.py:121-144)
+        for j, r in enumerate(range(self.first_block, self.last_block)):
+            if r == 0:
+                continue
+            Xr = self.X_i[j * w:(j + 1) * w]
+            Cr = self.C_i[j * w:(j + 1) * w]
+            be.spmm_block(self.A_ii[j], Xr.contiguous(), Cr, beta=0)
+            be.spmm_block(self.A_i0[j], self.X_0, Cr, beta=1)
+
+        # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155)
+        if self.first_block == 0 and self.n_owned > 0 and self.comm.rank == 0:
+            self.C_i[:w].copy_(self.C_0)
+
+    # -- result --------------------------------------------------------------
+
+    def allgather_result(self, C=None):
+        """All-gathers the per-rank result stripes into the full
+        (tiles_per_side*width, k) matrix (reference arrow_slim_mpi.py:415-425).
+        Returns numpy; fills C in place if given."""
+        w = self.width
+        k = self.C_i.shape[1]
+        pad_rows = self.blocks_per_rank * w
+        stripe = self.backend.zeros((pad_rows, k))
+        if self.n_owned:
+            stripe[:self.n_owned * w].copy_(self.C_i[:self.n_owned * w])
+        full = self.comm.allgather_cat(stripe)
+        out = full[:self.tiles_per_side * w]
+        out_np = out.cpu().numpy()
+        if C is not None:
+            C[:] = out_np
+            return C
+        return out_np

@@ -1,0 +1,127 @@
+"""Communication layer: RCCL over xGMI (torch.distributed backend "nccl")
+on GPU, gloo on CPU (tests), or a no-op single-process comm.
+
+Replaces the reference's mpi4py collectives (full call-site inventory in
+SURVEY.md §2): Bcast -> broadcast, Reduce(SUM) -> reduce, Alltoallv ->
+batched isend/irecv (RCCL has no alltoallv primitive; grouped ncclSend/Recv
+via torch.distributed.batch_isend_irecv is the native idiom),
+Gather+Bcast (allgather_result) -> all_gather, allreduce(LOR) -> all_reduce(MAX).
+
+One process per GPU; all payloads are torch tensors (CUDA tensors under
+nccl, CPU tensors under gloo — so the same engine code runs the gloo
+world_size>1 CPU tests and the RCCL GPU path).
+"""
+from typing import List, Optional, Sequence
+
+import torch
+
+try:
+    import torch.distributed as dist
+except Exception:  # pragma: no cover
+    dist = None
+
+
+class Comm:
+    """Single-process comm (world size 1): everything is local."""
+
+    rank = 0
+    size = 1
+
+    def barrier(self):
+        pass
+
+    def bcast_(self, tensor: torch.Tensor, src: int):
+        pass
+
+    def reduce_sum_(self, tensor: torch.Tensor, dst: int):
+        pass
+
+    def allreduce_max_(self, tensor: torch.Tensor):
+        pass
+
+    def alltoallv(self, send: torch.Tensor, send_counts: Sequence[int],
+                  recv_counts: Sequence[int]) -> torch.Tensor:
+        assert len(send_counts) == 1 and len(recv_counts) == 1
+        assert send_counts[0] == recv_counts[0]
+        return send
+
+    def allgather_cat(self, tensor: torch.Tensor) -> torch.Tensor:
+        return tensor
+
+
+class TorchDistComm(Comm):
+    """torch.distributed-backed comm (backend "nccl" == RCCL on ROCm, or
+    "gloo" for CPU tests)."""
+
+    def __init__(self, group=None):
+        assert dist is not None and dist.is_initialized(), \
+            "torch.distributed must be initialized"
+        self.group = group
+        self.rank = dist.get_rank(group)
+        self.size = dist.get_world_size(group)
+
+    def barrier(self):
+        dist.barrier(group=self.group)
+
+    def bcast_(self, tensor: torch.Tensor, src: int):
+        dist.broadcast(tensor, src=src, group=self.group)
+
+    def reduce_sum_(self, tensor: torch.Tensor, dst: int):
+        dist.reduce(tensor, dst=dst, op=dist.ReduceOp.SUM, group=self.group)
+
+    def allreduce_max_(self, tensor: torch.Tensor):
+        dist.all_reduce(tensor, op=dist.ReduceOp.MAX, group=self.group)
+
+    def alltoallv(self, send: torch.Tensor, send_counts: Sequence[int],
+                  recv_counts: Sequence[int]) -> torch.Tensor:
+        """Row-wise all-to-all-v: send rows [sum(send_counts[:r]) :
+        +send_counts[r]] to rank r; returns recv tensor with
+        sum(recv_counts) rows ordered by source rank.
+
+        Implemented as grouped point-to-point (batch_isend_irecv): the RCCL
+        alltoallv idiom over xGMI, and the only form gloo also supports.
+        The self-block is copied locally (never hits the backend).
+        """
+        assert send.dim() == 2
+        k = send.shape[1]
+        total_recv = int(sum(recv_counts))
+        recv = send.new_empty((total_recv, k))
+        sdispl = [0]
+        for c in send_counts:
+            sdispl.append(sdispl[-1] + int(c))
+        rdispl = [0]
+        for c in recv_counts:
+            rdispl.append(rdispl[-1] + int(c))
+
+        ops = []
+        for r in range(self.size):
+            if r == self.rank:
+                continue
+            if send_counts[r] > 0:
+                ops.append(dist.P2POp(dist.isend,
+                                      send[sdispl[r]:sdispl[r + 1]].contiguous(),
+                                      r, group=self.group))
+            if recv_counts[r] > 0:
+                ops.append(dist.P2POp(dist.irecv, recv[rdispl[r]:rdispl[r + 1]],
+                                      r, group=self.group))
+        if ops:
+            for w in dist.batch_isend_irecv(ops):
+                w.wait()
+        # local block
+        if send_counts[self.rank] > 0:
+            recv[rdispl[self.rank]:rdispl[self.rank + 1]].copy_(
+                send[sdispl[self.rank]:sdispl[self.rank] + send_counts[self.rank]])
+        return recv
+
+    def allgather_cat(self, tensor: torch.Tensor) -> torch.Tensor:
+        out: List[torch.Tensor] = [torch.empty_like(tensor) for _ in range(self.size)]
+        dist.all_gather(out, tensor.contiguous(), group=self.group)
+        return torch.cat(out, dim=0)
+
+
+def default_comm() -> Comm:
+    """TorchDistComm when torch.distributed is initialized, else the
+    single-process comm."""
+    if dist is not None and dist.is_available() and dist.is_initialized():
+        return TorchDistComm()
+    return Comm()

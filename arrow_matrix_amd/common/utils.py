@@ -1,0 +1,41 @@
+"""Utilities mirroring arrow/common/utils.py (str2bool:9-17, mpi_print:58-60,
+generate_sparse_matrix:63-87, generate_dense_matrix:90-99)."""
+import argparse
+from typing import Union
+
+import numpy as np
+from scipy import sparse
+
+
+def str2bool(v: Union[str, bool]) -> bool:
+    if isinstance(v, bool):
+        return v
+    if v.lower() in ('yes', 'true', 't', 'y', '1'):
+        return True
+    if v.lower() in ('no', 'false', 'f', 'n', '0'):
+        return False
+    raise argparse.ArgumentTypeError('Boolean value expected.')
+
+
+def mpi_print(rank: int, msg: str):
+    if rank == 0:
+        print(msg, flush=True)
+
+
+def generate_sparse_matrix(rows: int, cols: int, nnz: int, dtype,
+                           rng: np.random.Generator) -> sparse.csr_matrix:
+    """Fixed nonzeros-per-row random CSR (reference utils.py:63-87: nnz may
+    round up to a multiple of rows; duplicates summed)."""
+    nnzpr = int(np.ceil(nnz / rows))
+    actual_nnz = nnzpr * rows
+    data = rng.random((actual_nnz,), dtype=dtype)
+    indptr = np.arange(0, actual_nnz + 1, nnzpr, dtype=np.int64)
+    indices = rng.integers(0, cols, size=(actual_nnz,), dtype=np.int64)
+    tmp = sparse.csr_matrix((data, indices, indptr), shape=(rows, cols), dtype=dtype)
+    tmp.sum_duplicates()
+    tmp.sort_indices()
+    return tmp
+
+
+def generate_dense_matrix(rows: int, cols: int, dtype, rng: np.random.Generator) -> np.ndarray:
+    return 2 * rng.random((rows, cols), dtype=dtype) - 1

@@ -1,0 +1,163 @@
+"""On-disk decomposition format + block splitting.
+
+Re-implements (from scratch, semantics-identical) the reference's
+`arrow/common/graphio.py` `.npy`-CSR decomposition format so that decomposed
+arrow blocks produced by the reference drop in unchanged:
+
+- file naming:              graphio.py:38-70   (format_path)
+- save (new .npy format):   graphio.py:131-191 (save_decomposition_new)
+- load (new .npy format):   graphio.py:251-314 (load_decomposition_new;
+                            missing *_data.npy -> ones, graphio.py:297-298)
+- block splitting:          graphio.py:361-406 (split_matrix_to_blocks:
+                            keeps (0, i-1, i, i+1) blocks per row i>0, edge-pads
+                            indptr of a short last block-row to block_size and
+                            declares it square, graphio.py:389-399)
+
+No igraph dependency: the graph pickle of the reference is not read or
+written; only the CSR arrays + permutations (the part of the format the hot
+path consumes, arrow_dec_mpi.py:663).
+"""
+import enum
+import os
+from typing import List, Optional, Union
+
+import numpy as np
+from scipy import sparse
+
+
+class DecompositionFileType(enum.Enum):
+    npz = 1
+    indptr_npy = 2
+    indices_npy = 3
+    data_npy = 4
+    permutation_npy = 5
+    nonzero_rows_npy = 6
+
+
+_SUFFIX = {
+    DecompositionFileType.npz: ".npz",
+    DecompositionFileType.indptr_npy: "_indptr.npy",
+    DecompositionFileType.indices_npy: "_indices.npy",
+    DecompositionFileType.data_npy: "_data.npy",
+    DecompositionFileType.permutation_npy: "_permutation.npy",
+    DecompositionFileType.nonzero_rows_npy: "_nnzrows.npy",
+}
+
+
+def format_path(base_path: str, width: int, index: Optional[int], block_diagonal: bool,
+                file_type: DecompositionFileType) -> str:
+    """Same naming contract as reference graphio.py:38-70."""
+    path = f"{base_path}_B_{width}"
+    if index is not None:
+        path += f"_{index}"
+    if block_diagonal:
+        path += "_bd"
+    return path + _SUFFIX[file_type]
+
+
+def get_pathname(basename: str, width: int, is_block_diagonal: bool) -> str:
+    """Reference graphio.py:498-504."""
+    basename = f"{basename}_B"
+    if width:
+        basename += f"_{width}"
+    if is_block_diagonal:
+        basename += "_bd"
+    return basename
+
+
+def save_decomposition_new(decomposition, filename: str, width: int,
+                           block_diagonal: bool = True) -> None:
+    """Save parts as *_indptr/_indices/_data/_permutation .npy files.
+
+    `decomposition` is a list of (csr_matrix, permutation) pairs; `width` is
+    the decomposition's arrow width (shared across parts, as the reference
+    producer guarantees). File layout identical to reference
+    graphio.py:171-191 (the graph pickle / adjacency of the `save_graph`
+    branch is intentionally not written; the hot path never reads it).
+    """
+    for i, (B, permutation) in enumerate(decomposition):
+        B = sparse.csr_matrix(B)
+        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.indptr_npy), B.indptr)
+        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.indices_npy), B.indices)
+        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.data_npy), B.data)
+        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.permutation_npy),
+                np.asarray(permutation))
+
+
+def load_decomposition_new(filename: str, width: Optional[int] = None,
+                           block_diagonal: bool = True, no_permutation: bool = False,
+                           mem_map: bool = False):
+    """Load decomposition parts; stops at the first missing index.
+
+    Mirrors reference graphio.py:251-314: *_data.npy is optional (-> ones
+    float32, graphio.py:297-298). With mem_map=True the CSR is returned as the
+    raw (data, indices, indptr) tuple, as the reference does.
+    """
+    decomposition = []
+    i = 0
+    while True:
+        try:
+            f = format_path(filename, width, i, block_diagonal, DecompositionFileType.indptr_npy)
+            indptr = np.lib.format.open_memmap(f, mode='r') if mem_map else np.load(f)
+            f = format_path(filename, width, i, block_diagonal, DecompositionFileType.indices_npy)
+            indices = np.lib.format.open_memmap(f, mode='r') if mem_map else np.load(f)
+            f = format_path(filename, width, i, block_diagonal, DecompositionFileType.data_npy)
+            if os.path.exists(f):
+                data = np.lib.format.open_memmap(f, mode='r') if mem_map else np.load(f)
+            else:
+                data = np.ones(indices.size, dtype=np.float32)
+            if mem_map:
+                B = (data, indices, indptr)
+            else:
+                B = sparse.csr_matrix((data, indices, indptr))
+            if no_permutation:
+                permutation = None
+            else:
+                f = format_path(filename, width, i, block_diagonal, DecompositionFileType.permutation_npy)
+                permutation = np.load(f)
+        except FileNotFoundError:
+            break
+        decomposition.append((B, permutation))
+        i += 1
+    return decomposition
+
+
+def split_matrix_to_blocks(A: sparse.csr_matrix, block_size: int,
+                           dtype=None, use_min_shape: bool = False
+                           ) -> List[List[Union[sparse.csr_matrix, None]]]:
+    """Split A into the arrow-pattern blocks.
+
+    Semantics of reference graphio.py:361-406: for block-row i>0 only the
+    blocks j in {0, i-1, i, i+1} are materialised (None elsewhere); block-row 0
+    keeps every column block. A short last block-row is padded to block_size
+    rows by edge-padding indptr AND declared square (block_size x block_size).
+    Blocks are canonicalised (sum_duplicates + sort_indices).
+    """
+    rows, cols = A.shape
+    dtype = dtype or A.dtype
+
+    blocks_per_col = int(np.ceil(rows / block_size))
+    blocks_per_row = int(np.ceil(cols / block_size))
+    blocks: List[List[Union[sparse.csr_matrix, None]]] = \
+        [[None] * blocks_per_row for _ in range(blocks_per_col)]
+    for i in range(blocks_per_col):
+        for j in range(blocks_per_row):
+            if i > 0 and j not in (0, i - 1, i, i + 1):
+                continue
+            shape = (min(rows - i * block_size, block_size),
+                     min(cols - j * block_size, block_size))
+            sl = A[i * block_size:min(rows, (i + 1) * block_size),
+                   j * block_size:min(cols, (j + 1) * block_size)]
+            pad_width = block_size - shape[0]
+            if use_min_shape or pad_width == 0:
+                block = sparse.csr_matrix(sl, shape=shape, dtype=dtype)
+            else:
+                # Edge-pad indptr so the block has block_size rows
+                # (reference graphio.py:394-399).
+                indx_ptr = np.pad(sl.indptr, (0, pad_width), mode='edge')
+                block = sparse.csr_matrix((sl.data, sl.indices, indx_ptr),
+                                          shape=(block_size, block_size), dtype=dtype)
+            block.sum_duplicates()
+            block.sort_indices()
+            blocks[i][j] = block
+    return blocks

@@ -1,0 +1,121 @@
+"""Table construction parity: product (vectorised) vs oracle (literal
+restatement) vs golden vectors from the reference itself."""
+import os
+
+import numpy as np
+import pytest
+
+from arrow_matrix_amd import tables
+from oracle import all_to_all_tables_ref, aggregation_permutation_ref
+
+GOLDEN = os.path.join(os.path.dirname(__file__), 'golden', 'reference_katsets.npz')
+
+
+@pytest.fixture(scope='module')
+def golden():
+    return np.load(GOLDEN)
+
+
+def test_all_to_all_tables_vs_reference_golden(golden):
+    n = int(golden['a2a_n_cases'][0])
+    assert n >= 10
+    for ci in range(n):
+        perm = golden[f'a2a_{ci}_in_perm']
+        rpr, cols, total, off = (int(x) for x in golden[f'a2a_{ci}_meta'])
+        counts, displs, sp, ap = tables.all_to_all_tables(perm, rpr, cols, total, off)
+        np.testing.assert_array_equal(counts, golden[f'a2a_{ci}_counts'])
+        np.testing.assert_array_equal(displs, golden[f'a2a_{ci}_displs'])
+        np.testing.assert_array_equal(sp, golden[f'a2a_{ci}_send_perm'])
+        np.testing.assert_array_equal(ap, golden[f'a2a_{ci}_agg_perm'])
+
+
+def test_oracle_restatement_vs_reference_golden(golden):
+    n = int(golden['a2a_n_cases'][0])
+    for ci in range(n):
+        perm = golden[f'a2a_{ci}_in_perm']
+        rpr, cols, total, off = (int(x) for x in golden[f'a2a_{ci}_meta'])
+        counts, displs, sp, ap = all_to_all_tables_ref(perm, rpr, cols, total, off)
+        np.testing.assert_array_equal(counts, golden[f'a2a_{ci}_counts'])
+        np.testing.assert_array_equal(displs, golden[f'a2a_{ci}_displs'])
+        np.testing.assert_array_equal(sp, golden[f'a2a_{ci}_send_perm'])
+        np.testing.assert_array_equal(ap, golden[f'a2a_{ci}_agg_perm'])
+
+
+def test_product_vs_oracle_random():
+    rng = np.random.default_rng(3)
+    for _ in range(50):
+        rpr = int(rng.integers(1, 60))
+        total = int(rng.integers(1, 10))
+        off = int(rng.integers(0, total))
+        vals = rng.integers(0, rpr * (total + 2), size=rpr).astype(np.int64)
+        cols = int(rng.integers(1, 9))
+        a = tables.all_to_all_tables(vals, rpr, cols, total, off)
+        b = all_to_all_tables_ref(vals, rpr, cols, total, off)
+        for x, y in zip(a, b):
+            np.testing.assert_array_equal(np.asarray(x), np.asarray(y))
+
+
+def test_inverse_permutation_case():
+    """The reference's own unit test (test_arrowmpi.py:24-48)."""
+    ranks, prev_ranks, rpr, cols = 2, 6, 4, 6
+    permutation = np.asarray(list(reversed(range(ranks * rpr))))
+    for i in range(ranks):
+        sl = permutation[i * rpr:(i + 1) * rpr]
+        counts, displs, p, out_p = tables.all_to_all_tables(sl, rpr, cols,
+                                                            prev_ranks + ranks, prev_ranks)
+        assert counts[ranks + prev_ranks - i - 1] == rpr * cols
+        assert int(np.sum(counts)) == rpr * cols
+        assert displs[ranks + prev_ranks - i - 1] == 0
+        counts, displs, p, out_p = tables.all_to_all_tables(sl, rpr, cols,
+                                                            ranks + prev_ranks, 0)
+        assert counts[ranks - i - 1] == rpr * cols
+        assert displs[ranks - i - 1] == 0
+
+
+def test_routing_tables_match_reference_layout():
+    """At one block per rank, the generalised shared-rank tables reduce to the
+    reference's: same send grouping and receive order."""
+    rng = np.random.default_rng(5)
+    for _ in range(20):
+        nb_src = int(rng.integers(1, 6))
+        nb_dst = int(rng.integers(1, 6))
+        w = int(rng.integers(2, 10))
+        P = max(nb_src, nb_dst)
+        # a random to_next-style mapping from src rows onto dst rows (+ sentinel)
+        n_src, n_dst = nb_src * w, nb_dst * w
+        vals = rng.permutation(max(n_src, n_dst * 2))[:n_src].astype(np.int64)
+        vals[vals >= n_dst] = 2 * w * max(nb_src, nb_dst)  # sentinel
+        own_dst = tables.contiguous_block_owners(nb_dst, P)  # identity here
+        assert np.array_equal(own_dst, np.arange(nb_dst))
+        for r in range(nb_src):
+            sl = vals[r * w:(r + 1) * w]
+            # reference tables: total = nb_src + nb_dst, offset nb_src; its
+            # counts for dest rank nb_src+d == generalised counts for rank d
+            ref_counts, _, ref_sp, _ = all_to_all_tables_ref(sl, w, 1,
+                                                             nb_src + nb_dst, nb_src)
+            cnt, send_rows = tables.routing_send_tables(sl, w, own_dst, nb_dst, P)
+            np.testing.assert_array_equal(cnt[:nb_dst], np.asarray(ref_counts[nb_src:]))
+            n_valid = int(cnt.sum())
+            np.testing.assert_array_equal(send_rows, ref_sp[:n_valid])
+        own_src = tables.contiguous_block_owners(nb_src, P)
+        for r in range(nb_dst):
+            sl = vals_recv = rng.permutation(max(n_dst, n_src * 2))[:w].astype(np.int64)
+            sl = np.where(sl >= n_src, 2 * w * max(nb_src, nb_dst), sl)
+            ref_counts, _, _, ref_ap = all_to_all_tables_ref(sl, w, 1,
+                                                             nb_src + nb_dst, 0)
+            cnt, recv_rows = tables.routing_recv_tables(sl, w, own_src, nb_src, P)
+            np.testing.assert_array_equal(cnt[:nb_src], np.asarray(ref_counts[:nb_src]))
+            np.testing.assert_array_equal(recv_rows, ref_ap)
+
+
+def test_number_of_blocks_vs_golden(golden):
+    n = int(golden['nb_n_cases'][0])
+    for ci in range(n):
+        nnz_per_row = golden[f'nb_{ci}_nnz_per_row']
+        w = int(golden[f'nb_{ci}_width'][0])
+        assert tables.number_of_blocks(nnz_per_row, w) == int(golden[f'nb_{ci}_result'][0])
+
+
+def test_number_of_blocks_all_zero_raises():
+    with pytest.raises(ValueError):
+        tables.number_of_blocks(np.zeros(10, dtype=np.int64), 2)
