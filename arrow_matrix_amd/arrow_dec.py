@@ -106,27 +106,33 @@ class ArrowDecompositionMPI:
             own_r = tables.contiguous_block_owners(int(self.n_blocks[i + 1]), P)
             # forward (features i -> i+1): sender tables on my to_next slice
             # of matrix i, receiver tables on my to_prev slice of matrix i+1
+            # to_prev/to_next are ALREADY this rank's slices (one per part,
+            # as load_decomposition_new returns them — the reference's
+            # loader also hands each rank its own slice,
+            # arrow_dec_mpi.py:772-781)
+            tn_i = self._checked_slice(self._to_next[i], eng_s)
+            tp_n = self._checked_slice(self._to_prev[i + 1], eng_r)
             s_cnt, s_rows = tables.routing_send_tables(
-                self._slice_for_rank(self._to_next[i], eng_s), w, own_r,
-                int(self.n_blocks[i + 1]), P)
+                tn_i, w, own_r, int(self.n_blocks[i + 1]), P)
             r_cnt, r_rows = tables.routing_recv_tables(
-                self._slice_for_rank(self._to_prev[i + 1], eng_r), w, own_s,
-                int(self.n_blocks[i]), P)
+                tp_n, w, own_s, int(self.n_blocks[i]), P)
             self._forward[i] = _Exchange(s_cnt, s_rows, r_cnt, r_rows, eng_s.backend)
             # backward (partials i+1 -> i): sender tables on my to_prev slice
             # of matrix i+1, receiver tables on my to_next slice of matrix i
             s_cnt, s_rows = tables.routing_send_tables(
-                self._slice_for_rank(self._to_prev[i + 1], eng_r), w, own_s,
-                int(self.n_blocks[i]), P)
+                tp_n, w, own_s, int(self.n_blocks[i]), P)
             r_cnt, r_rows = tables.routing_recv_tables(
-                self._slice_for_rank(self._to_next[i], eng_s), w, own_r,
-                int(self.n_blocks[i + 1]), P)
+                tn_i, w, own_r, int(self.n_blocks[i + 1]), P)
             self._backward[i + 1] = _Exchange(s_cnt, s_rows, r_cnt, r_rows, eng_r.backend)
 
-    def _slice_for_rank(self, perm_full: np.ndarray, engine: ArrowSlimMPI) -> np.ndarray:
-        """This rank's rows of a matrix = its contiguous block span."""
+    def _checked_slice(self, perm_slice: np.ndarray, engine: ArrowSlimMPI) -> np.ndarray:
+        """Validate a per-rank permutation slice against the rank's span."""
         w = self.width
-        return perm_full[engine.first_block * w: engine.last_block * w]
+        expect = engine.n_owned * w
+        perm_slice = np.asarray(perm_slice)
+        assert perm_slice.size == expect, \
+            f"permutation slice has {perm_slice.size} rows, rank owns {expect}"
+        return perm_slice
 
     def load_data_from_blocks(self, blocked) -> None:
         """blocked: one block grid per decomposition part

@@ -62,6 +62,10 @@ class ArrowSlimMPI(ArrowMatrix):
         self.C_i: Optional[torch.Tensor] = None   # (n_owned*width, k) stripe
         self.C_0: Optional[torch.Tensor] = None   # (width, k) partial/reduced
         self.nnz_owned = 0
+        # optional HIP-event instrumentation of the SpMM kernel launches
+        # (filled by bench.py for the roofline; list of
+        # (start_event, end_event, nnz, rows) tuples)
+        self.kernel_events: Optional[list] = None
 
     # -- data loading --------------------------------------------------------
 
@@ -103,6 +107,21 @@ class ArrowSlimMPI(ArrowMatrix):
                 setattr(self, name, self.backend.zeros(shape))
             else:
                 buf.zero_()
+        # ping-pong pool: spmm writes C into a stripe that does NOT alias
+        # X_i, so `X := C` between iterations (set_features(result_tile()))
+        # is race-free on the GPU (the reference allocates a fresh C every
+        # iteration instead, arrow_slim_mpi.py:125-127)
+        self._stripe_bufs = [self.C_i, self.X_i]
+
+    def _select_result_buffer(self):
+        if self.X_i is None or self.C_i.data_ptr() != self.X_i.data_ptr():
+            return
+        for buf in self._stripe_bufs:
+            if buf.data_ptr() != self.X_i.data_ptr():
+                self.C_i = buf
+                return
+        self.C_i = self.backend.zeros(tuple(self.X_i.shape))
+        self._stripe_bufs.append(self.C_i)
 
     def set_features(self, X) -> None:
         """Stores a reference (no copy) when X already lives on this
@@ -136,6 +155,19 @@ class ArrowSlimMPI(ArrowMatrix):
         w = self.width
         k = self.X_i.shape[1]
 
+        def spmm_block(blk, Xr, Cr, beta):
+            if self.kernel_events is not None and be.device == 'cuda':
+                s = torch.cuda.Event(enable_timing=True)
+                e = torch.cuda.Event(enable_timing=True)
+                s.record()
+                be.spmm_block(blk, Xr, Cr, beta)
+                e.record()
+                self.kernel_events.append((s, e, blk.nnz, Cr.shape[0]))
+            else:
+                be.spmm_block(blk, Xr, Cr, beta)
+
+        self._select_result_buffer()
+
         # X_0 broadcast (arrow_slim_mpi.py:265-273); owner of block 0 is rank 0
         if self.first_block == 0 and self.n_owned > 0:
             self.X_0.copy_(self.X_i[:w])
@@ -145,7 +177,7 @@ class ArrowSlimMPI(ArrowMatrix):
         first = True
         for j, r in enumerate(range(self.first_block, self.last_block)):
             Xr = self.X_i[j * w:(j + 1) * w]
-            be.spmm_block(self.A_0i[j], Xr.contiguous(), self.C_0, beta=0 if first else 1)
+            spmm_block(self.A_0i[j], Xr.contiguous(), self.C_0, 0 if first else 1)
             first = False
         if first:  # rank owns no blocks of this matrix
             self.C_0.zero_()
@@ -159,8 +191,8 @@ class ArrowSlimMPI(ArrowMatrix):
                 continue
             Xr = self.X_i[j * w:(j + 1) * w]
             Cr = self.C_i[j * w:(j + 1) * w]
-            be.spmm_block(self.A_ii[j], Xr.contiguous(), Cr, beta=0)
-            be.spmm_block(self.A_i0[j], self.X_0, Cr, beta=1)
+            spmm_block(self.A_ii[j], Xr.contiguous(), Cr, 0)
+            spmm_block(self.A_i0[j], self.X_0, Cr, 1)
 
         # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155)
         if self.first_block == 0 and self.n_owned > 0 and self.comm.rank == 0:

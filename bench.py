@@ -1,0 +1,331 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: arrow-decomposition iterated SpMM on MI355X.
+
+Measures BASELINE.json's metric (iterated-SpMM GFLOP/s + achieved HBM GB/s)
+on the named configuration: at N=1 the workload is the configuration the
+metric is quoted on — the 100M-row synthetic planar-like graph, width 12.5M
+(8 block-rows), features k=128 (BASELINE.json configs[3]) — it fits one
+MI355X (~125 GB of 288 GB HBM3E). `--gpus N` shards the SAME matrix's
+block-rows over N ranks (strong scaling), launched by the driver as
+torchrun with one rank per GPU over RCCL.
+
+  python bench.py [--gpus N] [--steps K] [--warmup W] [--rows R] [--features k]
+
+Timing: W untimed warmup steps, then exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX over ranks; rank 0
+prints ONE JSON line. Inputs are resident in HBM before the timed region.
+
+The `roofline` object reports the dominant kernel (spmm_kernel): algorithmic
+bytes per launch (8*nnz + 4*(w+1) indptr/work-items + 4*k*w X-tile read +
+4*k*w C write — SURVEY.md §8d) divided by the HIP-event-measured average
+launch duration on the launch stream. `cpu_baseline` times the reference's
+own CPU arithmetic (scipy CSR @, arrow_slim_mpi.py:109-144 — restated via
+oracle semantics) on a bounded sample of the same workload on the host.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+if REPO not in sys.path:
+    sys.path.insert(0, REPO)
+
+
+def build_parser():
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=5)
+    p.add_argument('--warmup', type=int, default=2)
+    p.add_argument('--rows', type=int, default=100_000_000,
+                   help='total rows n (width = n / n-blocks)')
+    p.add_argument('--features', type=int, default=128)
+    p.add_argument('--n-blocks', type=int, default=8)
+    p.add_argument('--parts', type=int, default=1,
+                   help='decomposition length L (L>1 exercises the '
+                        'inter-part alltoallv exchange)')
+    p.add_argument('--device', type=str, default='gpu', choices=['gpu', 'cpu'])
+    p.add_argument('--band', type=int, default=1024,
+                   help='diagonal-block band half-width (planar-like locality)')
+    p.add_argument('--no-cpu-baseline', action='store_true')
+    p.add_argument('--json-out', type=str, default=None)
+    return p
+
+
+# --- synthetic planar-like arrow blocks (deterministic per block id) --------
+
+DIAG_DEG = 6
+COL_DEG = 2
+ROW0_DEG = 2
+HUB_ROWS = 16
+HUB_NNZ = 50_000
+VAL_SCALE = 0.55  # keeps ||X|| stable across iterations (sqrt(3/deg_total))
+
+
+def _gen_cols(role, w, deg, band, gen, device):
+    import torch
+    if role == 'diag':
+        base = torch.arange(w, device=device, dtype=torch.int64).unsqueeze(1)
+        delta = torch.randint(-band, band + 1, (w, deg), generator=gen, device=device)
+        cols = (base + delta) % w
+    else:
+        # quadratic skew toward low (hub) indices: planar-like graphs route
+        # long-range edges through high-degree vertices
+        u = torch.rand((w, deg), generator=gen, device=device)
+        cols = (u * u * w).long().clamp_(max=w - 1)
+    return torch.sort(cols, dim=1).values
+
+
+def generate_block(role, w, seed, device, band):
+    """Returns (indptr int64, indices int32, data f32) host numpy arrays for
+    one w x w block."""
+    import torch
+    deg = {'diag': DIAG_DEG, 'col': COL_DEG, 'row0': ROW0_DEG}[role]
+    gen = torch.Generator(device=device)
+    gen.manual_seed(seed)
+    cols = _gen_cols(role, w, deg, band, gen, device)
+    if role == 'row0' and w > 4 * HUB_NNZ:
+        # hub rows: the first rows of the arrow head are dense-ish
+        hub = torch.sort(
+            (torch.rand((HUB_ROWS, HUB_NNZ), generator=gen, device=device) * w)
+            .long().clamp_(max=w - 1), dim=1).values
+        counts = torch.full((w,), deg, dtype=torch.int64, device=device)
+        counts[:HUB_ROWS] += HUB_NNZ
+        indptr = torch.zeros(w + 1, dtype=torch.int64, device=device)
+        torch.cumsum(counts, 0, out=indptr[1:])
+        head = torch.cat([hub, cols[:HUB_ROWS]], dim=1)
+        head = torch.sort(head, dim=1).values
+        indices = torch.cat([head.flatten(), cols[HUB_ROWS:].flatten()])
+    else:
+        indices = cols.flatten()
+        indptr = torch.arange(0, indices.numel() + 1, deg, dtype=torch.int64,
+                              device=device)
+    nnz = indices.numel()
+    data = (torch.rand(nnz, generator=gen, device=device) * 2 - 1) * VAL_SCALE
+    return (indptr.cpu().numpy(),
+            indices.to(torch.int32).cpu().numpy(),
+            data.float().cpu().numpy())
+
+
+def block_csr(role, w, seed, device, band):
+    from scipy import sparse
+    indptr, indices, data = generate_block(role, w, seed, device, band)
+    return sparse.csr_matrix((data, indices, indptr), shape=(w, w))
+
+
+def expected_nnz(w, nb):
+    per_part = nb * (ROW0_DEG * w) + (nb - 1) * (DIAG_DEG + COL_DEG) * w
+    if w > 4 * HUB_NNZ:
+        per_part += nb * HUB_ROWS * HUB_NNZ
+    return per_part
+
+
+def build_blocks_for_rank(rank, world, w, nb, parts, gen_device, band):
+    """Block grids (only this rank's blocks materialised) for each part."""
+    grids = []
+    bpr = -(-nb // world)
+    first = min(rank * bpr, nb)
+    last = min(first + bpr, nb)
+    for p in range(parts):
+        grid = [[None] * nb for _ in range(nb)]
+        for c in range(first, last):
+            grid[0][c] = block_csr('row0', w, 10_000 * p + 10 * c + 1, gen_device, band)
+        for r in range(max(first, 1), last):
+            grid[r][r] = block_csr('diag', w, 10_000 * p + 10 * r + 2, gen_device, band)
+            grid[r][0] = block_csr('col', w, 10_000 * p + 10 * r + 3, gen_device, band)
+        grids.append(grid)
+    return grids, first, last
+
+
+def cpu_baseline_sample(w, band, k, threads):
+    """Time the reference CPU arithmetic (scipy CSR @ dense,
+    arrow_slim_mpi.py:109-144) on one diagonal block of the same workload."""
+    cap_w = min(w, 4_000_000)  # bound the sample to ~10-30 s of CPU work
+    A = block_csr('diag', cap_w, 2, 'cpu', band)
+    rng = np.random.default_rng(0)
+    X = (2 * rng.random((cap_w, k)) - 1).astype(np.float32)
+    t0 = time.perf_counter()
+    C = A @ X
+    t = time.perf_counter() - t0
+    del C
+    gflops = 2.0 * A.nnz * k / t / 1e9
+    return {
+        "value": round(gflops, 3), "unit": "GFLOP/s", "cores": threads,
+        "kind": "port",
+        "sample": f"one {cap_w}-row diagonal block ({A.nnz} nnz) x k={k}, "
+                  f"scipy CSR @ dense (the reference's cpu kernel), 1 iteration",
+    }
+
+
+def main():
+    args = build_parser().parse_args()
+    import torch
+
+    world = int(os.environ.get('WORLD_SIZE', '1'))
+    rank = int(os.environ.get('RANK', '0'))
+    local_rank = int(os.environ.get('LOCAL_RANK', str(rank)))
+    n_gpus = max(args.gpus, world)
+
+    use_gpu = args.device == 'gpu'
+    if use_gpu and not torch.cuda.is_available():
+        raise RuntimeError("bench.py --device gpu needs a HIP GPU "
+                           "(no silent CPU fallback); use --device cpu explicitly")
+
+    import torch.distributed as dist
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+    from arrow_matrix_amd.comm import TorchDistComm, Comm
+
+    if world > 1:
+        backend = 'nccl' if use_gpu else 'gloo'
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+        comm = TorchDistComm()
+    else:
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        comm = Comm()
+
+    nb = args.n_blocks
+    w = args.rows // nb
+    k = args.features
+    L = args.parts
+    gen_device = 'cuda' if use_gpu else 'cpu'
+
+    t0 = time.perf_counter()
+    grids, first, last = build_blocks_for_rank(comm.rank, comm.size, w, nb, L,
+                                               gen_device, args.band)
+    if rank == 0:
+        print(f"# generated blocks in {time.perf_counter()-t0:.1f}s "
+              f"(rank owns block-rows [{first},{last}))", file=sys.stderr)
+
+    n_blocks = np.full(L, nb, dtype=np.int64)
+    if L > 1:
+        # identity permutations between parts -> local-heavy exchange, but the
+        # full alltoallv path runs
+        to_prev = [None] + [np.arange(nb * w, dtype=np.int64)[first * w:last * w]] * (L - 1)
+        to_next = [np.arange(nb * w, dtype=np.int64)[first * w:last * w]] * (L - 1) + [None]
+    else:
+        to_prev, to_next = [None], [None]
+
+    arrow = ArrowDecompositionMPI.initialize(comm, n_blocks, to_prev, to_next,
+                                             w, k, device=args.device)
+    t0 = time.perf_counter()
+    arrow.load_data_from_blocks(grids)
+    del grids
+    arrow.zero_rhs(w, k)
+    if rank == 0:
+        print(f"# uploaded blocks in {time.perf_counter()-t0:.1f}s", file=sys.stderr)
+
+    # features resident on device before the timed region
+    eng0 = arrow.engines[0]
+    if use_gpu:
+        g = torch.Generator(device='cuda')
+        g.manual_seed(42 + comm.rank)
+        X = torch.rand(eng0.X_i.shape, generator=g, device='cuda') * 2 - 1
+    else:
+        X = torch.rand(eng0.X_i.shape) * 2 - 1
+    eng0.set_features(X)
+
+    def sync():
+        comm.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    def one_step():
+        arrow.step()
+        eng0.set_features(eng0.result_tile())
+
+    for _ in range(args.warmup):
+        one_step()
+    sync()
+
+    # instrument the SpMM kernel launches for the roofline
+    events = []
+    for eng in arrow.engines:
+        eng.kernel_events = events
+
+    sync()
+    t_start = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    sync()
+    elapsed = time.perf_counter() - t_start
+
+    for eng in arrow.engines:
+        eng.kernel_events = None
+
+    # MAX over ranks
+    t_t = torch.tensor([elapsed], dtype=torch.float64,
+                       device='cuda' if (world > 1 and use_gpu) else 'cpu')
+    comm.allreduce_max_(t_t)
+    elapsed = float(t_t.item())
+
+    nnz_total = expected_nnz(w, nb) * L
+    flops = 2.0 * nnz_total * k * args.steps
+    gflops = flops / elapsed / 1e9
+
+    # roofline from the kernel events (this rank; rank 0 reports)
+    roofline = None
+    if use_gpu and events:
+        total_ms = 0.0
+        total_bytes = 0.0
+        for s, e, nnz, rows in events:
+            total_ms += s.elapsed_time(e)
+            total_bytes += 8.0 * nnz + 4.0 * (rows + 1) + 8.0 * k * rows
+        achieved = total_bytes / (total_ms / 1e3) / 1e9  # GB/s
+        peak = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(achieved, 1),
+            "peak": peak,
+            "unit": "GB/s",
+            "frac": round(achieved / peak, 4),
+            "traffic": None,
+            "kernel": "spmm_kernel",
+            "launches": len(events),
+            "avg_launch_ms": round(total_ms / len(events), 4),
+        }
+
+    cpu_base = None
+    if rank == 0 and world <= 1 and not args.no_cpu_baseline:
+        cpu_base = cpu_baseline_sample(w, args.band, k, threads=1)
+
+    if rank == 0:
+        result = {
+            "metric": "iterated_spmm_gflops",
+            "value": round(gflops, 2),
+            "unit": "GFLOP/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "f32",
+            "data": "synthetic",
+            "config": {
+                "workload": f"cfg4_synth_{args.rows//1_000_000}M_k{k}",
+                "rows": args.rows, "width": w, "n_blocks": nb, "parts": L,
+                "features": k, "nnz": nnz_total, "band": args.band,
+                "parallelism": f"block-rows over {n_gpus} GPU(s), RCCL/xGMI",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_base,
+        }
+        line = json.dumps(result)
+        print(line)
+        if args.json_out:
+            with open(args.json_out, 'w') as f:
+                f.write(line + "\n")
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

@@ -1,0 +1,135 @@
+"""HIP kernel parity vs the scipy oracle arithmetic — runs on a real MI355X
+(`pytest -m gpu`). Covers the reference's operating points (k = 16/32/128,
+arrow_bench defaults) plus edge cases: empty matrices, empty rows, ragged k,
+dense hub rows (the long-row atomic path), k > 256 column tiling,
+beta accumulate."""
+import numpy as np
+import pytest
+import torch
+from scipy import sparse
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope='module')
+def gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from arrow_matrix_amd import hip
+    hip.set_device(torch.cuda.current_device())
+    return hip
+
+
+def _random_csr(rows, cols, density, seed, dtype=np.float32):
+    rs = np.random.RandomState(seed)
+    m = sparse.random(rows, cols, density=density, format='csr', random_state=rs,
+                      dtype=np.float64)
+    return sparse.csr_matrix(m, dtype=dtype)
+
+
+def _check_spmm(gpu, A, k, beta, seed=0, rtol=1e-5, atol=1e-5):
+    rng = np.random.default_rng(seed)
+    X = (2 * rng.random((A.shape[1], k)) - 1).astype(np.float32)
+    C0 = (2 * rng.random((A.shape[0], k)) - 1).astype(np.float32)
+    Xt = torch.from_numpy(X).cuda()
+    Ct = torch.from_numpy(C0.copy()).cuda()
+    blk = gpu.CsrBlockGPU(A)
+    blk.spmm(Xt.data_ptr(), Ct.data_ptr(), k, beta,
+             torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    ref = (A @ X) + (C0 if beta else 0)
+    got = Ct.cpu().numpy()
+    scale = max(1.0, float(np.abs(ref).max()))
+    np.testing.assert_allclose(got, ref, rtol=rtol, atol=atol * scale)
+
+
+@pytest.mark.parametrize("k", [1, 2, 3, 4, 5, 16, 32, 64, 128])
+def test_spmm_k_sweep(gpu, k):
+    A = _random_csr(500, 700, 0.02, seed=k)
+    _check_spmm(gpu, A, k, beta=0, seed=k)
+
+
+@pytest.mark.parametrize("k", [16, 128])
+def test_spmm_accumulate(gpu, k):
+    A = _random_csr(300, 300, 0.05, seed=40 + k)
+    _check_spmm(gpu, A, k, beta=1, seed=k)
+
+
+def test_spmm_k_above_column_tile(gpu):
+    # k > 256 exercises the column-offset loop; 260 also the guard
+    for k in (260, 512):
+        A = _random_csr(100, 120, 0.05, seed=k)
+        _check_spmm(gpu, A, k, beta=0, seed=k)
+
+
+def test_spmm_empty_matrix(gpu):
+    A = sparse.csr_matrix((64, 64), dtype=np.float32)
+    _check_spmm(gpu, A, 16, beta=0)   # beta=0 must zero all rows
+    _check_spmm(gpu, A, 16, beta=1)   # beta=1 must leave C unchanged
+
+
+def test_spmm_empty_rows_beta0_zeroes(gpu):
+    A = _random_csr(200, 200, 0.02, seed=3).tolil()
+    A[50:100] = 0
+    A = sparse.csr_matrix(A, dtype=np.float32)
+    _check_spmm(gpu, A, 32, beta=0)
+
+
+def test_spmm_dense_hub_rows_long_row_split(gpu):
+    """Rows with nnz >> SEG_NNZ take the segmented atomic path."""
+    rows, cols = 64, 20000
+    A = _random_csr(rows, cols, 0.001, seed=5).tolil()
+    rs = np.random.RandomState(6)
+    for r in (0, 13):
+        A[r] = rs.rand(cols) * (rs.rand(cols) < 0.6)  # ~12000 nnz
+    A = sparse.csr_matrix(A, dtype=np.float32)
+    assert int(A.getnnz(1).max()) > 2048
+    for beta in (0, 1):
+        _check_spmm(gpu, A, 128, beta=beta, rtol=1e-4, atol=1e-4)
+
+
+def test_spmm_single_row_single_col(gpu):
+    A = sparse.csr_matrix(np.array([[2.5]], dtype=np.float32))
+    _check_spmm(gpu, A, 7, beta=0)
+
+
+def test_gather_scatter_roundtrip(gpu):
+    rng = np.random.default_rng(0)
+    for k in (4, 16, 33, 128):
+        n = 500
+        src = torch.from_numpy(rng.random((n, k), dtype=np.float32)).cuda()
+        idx_np = rng.permutation(n).astype(np.int64)
+        idx = torch.from_numpy(idx_np).cuda()
+        dst = torch.empty_like(src)
+        s = torch.cuda.current_stream().cuda_stream
+        gpu.gather_rows(src.data_ptr(), dst.data_ptr(), idx.data_ptr(), n, k, s)
+        torch.cuda.synchronize()
+        np.testing.assert_array_equal(dst.cpu().numpy(), src.cpu().numpy()[idx_np])
+        # scatter back inverts
+        dst2 = torch.empty_like(src)
+        gpu.scatter_rows(dst2.data_ptr(), dst.data_ptr(), idx.data_ptr(), n, k, s)
+        torch.cuda.synchronize()
+        np.testing.assert_array_equal(dst2.cpu().numpy(), src.cpu().numpy())
+
+
+def test_scatter_add(gpu):
+    rng = np.random.default_rng(1)
+    n, m, k = 300, 200, 16
+    src = rng.random((n, k), dtype=np.float32)
+    base = rng.random((m, k), dtype=np.float32)
+    idx_np = rng.integers(0, m, size=n).astype(np.int64)  # duplicate targets? no:
+    # duplicates in idx would race in the kernel unless atomics; the routing
+    # use-case (receive permutation) is duplicate-free — enforce that here
+    idx_np = rng.permutation(m)[:min(n, m)].astype(np.int64)
+    src = src[:idx_np.size]
+    dst = torch.from_numpy(base.copy()).cuda()
+    s = torch.cuda.current_stream().cuda_stream
+    srct = torch.from_numpy(src).cuda()
+    idxt = torch.from_numpy(idx_np).cuda()
+    from arrow_matrix_amd import hip
+    hip.scatter_add_rows(dst.data_ptr(), srct.data_ptr(), idxt.data_ptr(),
+                         idx_np.size, k, s)
+    torch.cuda.synchronize()
+    ref = base.copy()
+    ref[idx_np] += src
+    np.testing.assert_allclose(dst.cpu().numpy(), ref, rtol=1e-6, atol=1e-6)
