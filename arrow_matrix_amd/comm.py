@@ -48,6 +48,13 @@ class Comm:
     def allgather_cat(self, tensor: torch.Tensor) -> torch.Tensor:
         return tensor
 
+    def allgather_int(self, value: int) -> List[int]:
+        return [int(value)]
+
+    def alltoall_ints(self, values: Sequence[int]) -> List[int]:
+        assert len(values) == 1
+        return [int(values[0])]
+
 
 class TorchDistComm(Comm):
     """torch.distributed-backed comm (backend "nccl" == RCCL on ROCm, or
@@ -117,6 +124,17 @@ class TorchDistComm(Comm):
         out: List[torch.Tensor] = [torch.empty_like(tensor) for _ in range(self.size)]
         dist.all_gather(out, tensor.contiguous(), group=self.group)
         return torch.cat(out, dim=0)
+
+    def allgather_int(self, value: int) -> List[int]:
+        t = torch.tensor([int(value)], dtype=torch.int64)
+        return [int(x.item()) for x in self.allgather_cat(t)]
+
+    def alltoall_ints(self, values: Sequence[int]) -> List[int]:
+        """MPI Alltoall of one int per peer (matrix_slice.py:248)."""
+        assert len(values) == self.size
+        send = torch.tensor(values, dtype=torch.int64).view(-1, 1)
+        recv = self.alltoallv(send, [1] * self.size, [1] * self.size)
+        return [int(x.item()) for x in recv.view(-1)]
 
 
 def default_comm() -> Comm:
