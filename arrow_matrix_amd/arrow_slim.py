@@ -62,35 +62,112 @@ class ArrowSlimMPI(ArrowMatrix):
         self.C_i: Optional[torch.Tensor] = None   # (n_owned*width, k) stripe
         self.C_0: Optional[torch.Tensor] = None   # (width, k) partial/reduced
         self.nnz_owned = 0
+        # merged resident GPU structures (built at load time, gpu only)
+        self._A_row0 = None
+        self._A_rest = None
+        self._rest_row_offset = 0
         # optional HIP-event instrumentation of the SpMM kernel launches
         # (filled by bench.py for the roofline; list of
-        # (start_event, end_event, nnz, rows) tuples)
+        # (start_event, end_event, nnz, c_rows, x_rows) tuples)
         self.kernel_events: Optional[list] = None
 
     # -- data loading --------------------------------------------------------
 
     def load_sparse_matrix_from_blocks(self, blocks) -> None:
         """blocks: block grid (list of lists; entries not owned by this rank
-        may be None). Owned pieces are uploaded once and stay resident."""
+        may be None). Owned pieces are uploaded once and stay resident.
+
+        GPU path: the owned blocks are MERGED into two resident structures
+         so each iteration is two fused launches
+        instead of 3·blocks:
+          * `_A_row0`: all owned A_0c side by side (w x n_owned·w), read
+            against the X stripe -> C_0 written once;
+          * `_A_rest`: per-row merge of A_rr (columns -> stripe) and A_r0
+            (columns negative-encoded -> X_0) for owned r > 0 -> C written
+            once per row (the reference multiplies and accumulates
+            separately, arrow_slim_mpi.py:121-144).
+        CPU path keeps per-block scipy (the reference's own cpu dataflow).
+        """
         assert len(blocks) == self.tiles_per_side
         self.A_0i, self.A_ii, self.A_i0 = [], [], []
         self.nnz_owned = 0
+        gpu = self.backend.device == 'cuda'
         for r in range(self.first_block, self.last_block):
             b0r = blocks[0][r]
             assert b0r is not None, f"missing block (0,{r})"
             if self.width is None:
                 self.width = b0r.shape[0]
-            self.A_0i.append(self.backend.upload_block(b0r))
+            self.A_0i.append(b0r if gpu else self.backend.upload_block(b0r))
             self.nnz_owned += b0r.nnz
             if r > 0:
                 assert blocks[r][r] is not None and blocks[r][0] is not None, \
                     f"missing diagonal/column block for row {r}"
-                self.A_ii.append(self.backend.upload_block(blocks[r][r]))
-                self.A_i0.append(self.backend.upload_block(blocks[r][0]))
-                self.nnz_owned += blocks[r][r].nnz + blocks[r][0].nnz
+                rr, r0 = blocks[r][r], blocks[r][0]
+                self.A_ii.append(rr if gpu else self.backend.upload_block(rr))
+                self.A_i0.append(r0 if gpu else self.backend.upload_block(r0))
+                self.nnz_owned += rr.nnz + r0.nnz
             else:
                 self.A_ii.append(None)
                 self.A_i0.append(None)
+        if gpu:
+            self._build_merged_gpu()
+            # drop the host block references — device copies are resident
+            self.A_0i = self.A_ii = self.A_i0 = None
+
+    def _build_merged_gpu(self) -> None:
+        w = self.width
+        nw = self.n_owned
+        self._A_row0 = None
+        self._A_rest = None
+        self._rest_row_offset = 0
+        if nw == 0:
+            return
+        # --- row-0 merge: C_0 = [A_0,first .. A_0,last-1] @ X_stripe -------
+        rows_cat, cols_cat, data_cat = [], [], []
+        for j, b in enumerate(self.A_0i):
+            b = b.tocsr()
+            rows_cat.append(np.repeat(np.arange(w), np.diff(b.indptr)))
+            cols_cat.append(b.indices.astype(np.int64) + j * w)
+            data_cat.append(b.data.astype(np.float32))
+        self._A_row0 = self._merged_handle(w, nw * w, rows_cat, cols_cat,
+                                           data_cat, x_rows=nw * w)
+        # --- rest merge: C[r] = A_rr @ X_r  +  A_r0 @ X_0 ------------------
+        self._rest_row_offset = w if self.first_block == 0 else 0
+        rows_cat, cols_cat, data_cat = [], [], []
+        n_rest = 0
+        for j, r in enumerate(range(self.first_block, self.last_block)):
+            if r == 0:
+                continue
+            rr = self.A_ii[j].tocsr()
+            r0 = self.A_i0[j].tocsr()
+            local = j * w - self._rest_row_offset
+            rows_cat.append(local + np.repeat(np.arange(w), np.diff(rr.indptr)))
+            cols_cat.append(rr.indices.astype(np.int64) + j * w)
+            data_cat.append(rr.data.astype(np.float32))
+            rows_cat.append(local + np.repeat(np.arange(w), np.diff(r0.indptr)))
+            cols_cat.append(-(r0.indices.astype(np.int64) + 1))  # -> X_0
+            data_cat.append(r0.data.astype(np.float32))
+            n_rest += 1
+        if n_rest:
+            rest_rows = nw * w - self._rest_row_offset
+            self._A_rest = self._merged_handle(rest_rows, nw * w, rows_cat,
+                                               cols_cat, data_cat,
+                                               x_rows=n_rest * w + w)
+
+    def _merged_handle(self, n_rows, n_cols, rows_cat, cols_cat, data_cat,
+                       x_rows):
+        rows = np.concatenate(rows_cat)
+        cols = np.concatenate(cols_cat)
+        data = np.concatenate(data_cat)
+        order = np.argsort(rows, kind='stable')
+        rows, cols, data = rows[order], cols[order], data[order]
+        indptr = np.zeros(n_rows + 1, dtype=np.int64)
+        np.cumsum(np.bincount(rows, minlength=n_rows), out=indptr[1:])
+        handle = self.backend.upload_arrays((n_rows, n_cols), indptr,
+                                            cols.astype(np.int32),
+                                            data.astype(np.float32))
+        handle.x_rows = x_rows  # algorithmic X rows for the roofline
+        return handle
 
     # -- buffers -------------------------------------------------------------
 
@@ -151,52 +228,86 @@ class ArrowSlimMPI(ArrowMatrix):
         if device is not None and device != self.device:
             raise ValueError(
                 f"engine was built for device={self.device!r}; got {device!r}")
-        be = self.backend
-        w = self.width
-        k = self.X_i.shape[1]
-
-        def spmm_block(blk, Xr, Cr, beta):
-            if self.kernel_events is not None and be.device == 'cuda':
-                s = torch.cuda.Event(enable_timing=True)
-                e = torch.cuda.Event(enable_timing=True)
-                s.record()
-                be.spmm_block(blk, Xr, Cr, beta)
-                e.record()
-                self.kernel_events.append((s, e, blk.nnz, Cr.shape[0]))
-            else:
-                be.spmm_block(blk, Xr, Cr, beta)
-
         self._select_result_buffer()
+        w = self.width
 
         # X_0 broadcast (arrow_slim_mpi.py:265-273); owner of block 0 is rank 0
         if self.first_block == 0 and self.n_owned > 0:
             self.X_0.copy_(self.X_i[:w])
         self.comm.bcast_(self.X_0, src=0)
 
-        # C_0 partial: on-GPU accumulation over owned first-block-row tiles
-        first = True
-        for j, r in enumerate(range(self.first_block, self.last_block)):
-            Xr = self.X_i[j * w:(j + 1) * w]
-            spmm_block(self.A_0i[j], Xr.contiguous(), self.C_0, 0 if first else 1)
-            first = False
-        if first:  # rank owns no blocks of this matrix
+        if self.backend.device == 'cuda':
+            self._spmm_gpu()
+        else:
+            self._spmm_cpu()
+
+        # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155)
+        if self.first_block == 0 and self.n_owned > 0 and self.comm.rank == 0:
+            self.C_i[:w].copy_(self.C_0)
+
+    def _timed(self, fn, nnz, c_rows, x_rows):
+        """HIP-event instrumentation for the roofline (bench.py)."""
+        if self.kernel_events is None:
+            fn()
+            return
+        s = torch.cuda.Event(enable_timing=True)
+        e = torch.cuda.Event(enable_timing=True)
+        s.record()
+        fn()
+        e.record()
+        self.kernel_events.append((s, e, nnz, c_rows, x_rows))
+
+    def _spmm_gpu(self) -> None:
+        """Two fused launches on the resident merged structures
+        (DESIGN.md §kernels)."""
+        be = self.backend
+        w = self.width
+
+        # C_0 = A_row0_merged @ X_stripe  (ONE launch; the reference runs one
+        # CSRMM per block and re-uploads A and X, arrow_slim_mpi.py:181-195)
+        if self._A_row0 is not None:
+            h = self._A_row0
+            self._timed(lambda: be.spmm_block(h, self.X_i, self.C_0, 0),
+                        h.nnz, w, h.x_rows)
+        else:
             self.C_0.zero_()
 
         # ONE reduce of the first block-row partials (arrow_slim_mpi.py:116)
         self.comm.reduce_sum_(self.C_0, dst=0)
 
-        # Diagonal + first-block-column tiles (arrow_slim_mpi.py:121-144)
+        # C_rest = A_diag_merged @ X_stripe + A_col_merged @ X_0 fused:
+        # C written once (vs the reference's multiply-then-accumulate,
+        # arrow_slim_mpi.py:121-144)
+        if self._A_rest is not None:
+            h = self._A_rest
+            C_sub = self.C_i[self._rest_row_offset:
+                             self.n_owned * w]
+            self._timed(lambda: be.spmm_dual(h, self.X_i, self.X_0, C_sub, 0),
+                        h.nnz, C_sub.shape[0], h.x_rows)
+
+    def _spmm_cpu(self) -> None:
+        """Per-block scipy dataflow — the reference's own cpu path
+        (arrow_slim_mpi.py:78-156)."""
+        be = self.backend
+        w = self.width
+
+        first = True
+        for j, r in enumerate(range(self.first_block, self.last_block)):
+            Xr = self.X_i[j * w:(j + 1) * w]
+            be.spmm_block(self.A_0i[j], Xr.contiguous(), self.C_0, 0 if first else 1)
+            first = False
+        if first:  # rank owns no blocks of this matrix
+            self.C_0.zero_()
+
+        self.comm.reduce_sum_(self.C_0, dst=0)
+
         for j, r in enumerate(range(self.first_block, self.last_block)):
             if r == 0:
                 continue
             Xr = self.X_i[j * w:(j + 1) * w]
             Cr = self.C_i[j * w:(j + 1) * w]
-            spmm_block(self.A_ii[j], Xr.contiguous(), Cr, 0)
-            spmm_block(self.A_i0[j], self.X_0, Cr, 1)
-
-        # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155)
-        if self.first_block == 0 and self.n_owned > 0 and self.comm.rank == 0:
-            self.C_i[:w].copy_(self.C_0)
+            be.spmm_block(self.A_ii[j], Xr.contiguous(), Cr, 0)
+            be.spmm_block(self.A_i0[j], self.X_0, Cr, 1)
 
     # -- result --------------------------------------------------------------
 

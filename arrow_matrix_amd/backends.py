@@ -92,6 +92,15 @@ class GpuBackend:
         assert X.is_contiguous() and C.is_contiguous()
         block.spmm(X.data_ptr(), C.data_ptr(), C.shape[1], beta, self._stream())
 
+    def upload_arrays(self, shape, indptr, indices, data) -> hip.CsrBlockGPU:
+        return hip.CsrBlockGPU(arrays=(shape, indptr, indices, data))
+
+    def spmm_dual(self, block: hip.CsrBlockGPU, X0: torch.Tensor,
+                  X1: torch.Tensor, C: torch.Tensor, beta: int):
+        assert X0.is_contiguous() and X1.is_contiguous() and C.is_contiguous()
+        block.spmm_dual(X0.data_ptr(), X1.data_ptr(), C.data_ptr(),
+                        C.shape[1], beta, self._stream())
+
     def gather_rows(self, src: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
         out = torch.empty((idx.shape[0], src.shape[1]), dtype=torch.float32,
                           device=self.torch_device)

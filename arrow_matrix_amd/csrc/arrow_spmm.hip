@@ -55,9 +55,11 @@ constexpr int BLOCK_THREADS = 256;
 
 struct CsrBlock {
   int64_t rows = 0, cols = 0, nnz = 0;
-  int32_t *indices = nullptr;   // device, nnz
-  float *data = nullptr;        // device, nnz
-  // work items (device): row (bit31 = atomic), [begin, end) into indices/data
+  // (col, val) packed 8-byte pairs (device, nnz). col < 0 encodes an index
+  // into the SECOND feature matrix of arrow_spmm_dual: col = -(idx+1)
+  // (the fused diagonal+first-block-column layout).
+  int2 *pairs = nullptr;
+  // work items (device): row (bit31 = atomic), [begin, end) into pairs
   int32_t *item_row = nullptr;
   int32_t *item_begin = nullptr;
   int32_t *item_end = nullptr;
@@ -76,11 +78,11 @@ int64_t g_next_handle = 1;
 
 template <int VEC, int GROUP, int BETA, bool GUARD>
 __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
-    const int32_t *__restrict__ indices, const float *__restrict__ data,
+    const int2 *__restrict__ pairs,
     const int32_t *__restrict__ item_row, const int32_t *__restrict__ item_begin,
     const int32_t *__restrict__ item_end, int64_t n_items,
-    const float *__restrict__ X, float *__restrict__ C, int64_t k,
-    int64_t col_off) {
+    const float *__restrict__ X0, const float *__restrict__ X1,
+    float *__restrict__ C, int64_t k, int64_t col_off) {
   constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
   const int lane_in_group = threadIdx.x % GROUP;
   const int group_in_block = threadIdx.x / GROUP;
@@ -101,11 +103,17 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
 #pragma unroll
     for (int j = 0; j < VEC; ++j) acc[j] = 0.0f;
 
-    if (active) {
-      for (int32_t t = b; t < e; ++t) {
-        const int32_t c = indices[t];
-        const float v = data[t];
-        const float *xr = X + (int64_t)c * k + col0;
+    // Stage GROUP (col,val) pairs cooperatively (ONE coalesced 8-B/lane
+    // load per GROUP nonzeros), then broadcast them across the group with
+    // wave shuffles — the A stream costs 1/GROUP memory instructions per
+    // nonzero instead of 2, and the X-row loads below can be issued back
+    // to back by the unrolled loop.
+    auto consume = [&](int2 mine, int u) {
+      const int32_t c = __shfl(mine.x, u, GROUP);
+      const float v = __int_as_float(__shfl(mine.y, u, GROUP));
+      const float *xr = (c < 0 ? X1 + (int64_t)(-c - 1) * k
+                               : X0 + (int64_t)c * k) + col0;
+      if (active) {
         if constexpr (VEC == 4) {
           const float4 xv = *reinterpret_cast<const float4 *>(xr);
           acc[0] = fmaf(v, xv.x, acc[0]);
@@ -120,6 +128,23 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
           acc[0] = fmaf(v, xr[0], acc[0]);
         }
       }
+    };
+    // full GROUP-sized chunks: constant trip count, partially unrolled so
+    // several X-row loads are in flight per group
+    int32_t base = b;
+    for (; base + GROUP <= e; base += GROUP) {
+      const int2 mine = pairs[base + lane_in_group];
+#pragma unroll 8
+      for (int u = 0; u < GROUP; ++u) consume(mine, u);
+    }
+    if (base < e) {  // remainder (< GROUP entries)
+      const int32_t t = base + lane_in_group;
+      const int2 mine = (t < e) ? pairs[t] : int2{0, 0};
+      const int cnt = e - base;
+      for (int u = 0; u < cnt; ++u) consume(mine, u);
+    }
+
+    if (active) {
       float *cr = C + (int64_t)row * k + col0;
       if (is_split) {
 #pragma unroll
@@ -242,8 +267,8 @@ LaunchCfg pick_cfg(int64_t k) {
 }
 
 template <int VEC, int GROUP>
-int launch_spmm_vg(const CsrBlock &blk, const float *X, float *C, int64_t k,
-                   int beta, hipStream_t stream) {
+int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
+                   float *C, int64_t k, int beta, hipStream_t stream) {
   constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
   int blocks = (int)std::min<int64_t>(
       (blk.n_items + GROUPS_PER_BLOCK - 1) / GROUPS_PER_BLOCK, 8192);
@@ -253,8 +278,8 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X, float *C, int64_t k,
     const bool guard = (col_off + span > k);
     auto run = [&](auto kern) {
       hipLaunchKernelGGL(kern, dim3(blocks), dim3(BLOCK_THREADS), 0, stream,
-                         blk.indices, blk.data, blk.item_row, blk.item_begin,
-                         blk.item_end, blk.n_items, X, C, k, col_off);
+                         blk.pairs, blk.item_row, blk.item_begin,
+                         blk.item_end, blk.n_items, X0, X1, C, k, col_off);
     };
     if (beta == 0) {
       if (guard) run(spmm_kernel<VEC, GROUP, 0, true>);
@@ -269,16 +294,16 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X, float *C, int64_t k,
 }
 
 template <int VEC>
-int launch_spmm_v(const CsrBlock &blk, const float *X, float *C, int64_t k,
-                  int beta, int group, hipStream_t stream) {
+int launch_spmm_v(const CsrBlock &blk, const float *X0, const float *X1,
+                  float *C, int64_t k, int beta, int group, hipStream_t stream) {
   switch (group) {
-    case 1:  return launch_spmm_vg<VEC, 1>(blk, X, C, k, beta, stream);
-    case 2:  return launch_spmm_vg<VEC, 2>(blk, X, C, k, beta, stream);
-    case 4:  return launch_spmm_vg<VEC, 4>(blk, X, C, k, beta, stream);
-    case 8:  return launch_spmm_vg<VEC, 8>(blk, X, C, k, beta, stream);
-    case 16: return launch_spmm_vg<VEC, 16>(blk, X, C, k, beta, stream);
-    case 32: return launch_spmm_vg<VEC, 32>(blk, X, C, k, beta, stream);
-    default: return launch_spmm_vg<VEC, 64>(blk, X, C, k, beta, stream);
+    case 1:  return launch_spmm_vg<VEC, 1>(blk, X0, X1, C, k, beta, stream);
+    case 2:  return launch_spmm_vg<VEC, 2>(blk, X0, X1, C, k, beta, stream);
+    case 4:  return launch_spmm_vg<VEC, 4>(blk, X0, X1, C, k, beta, stream);
+    case 8:  return launch_spmm_vg<VEC, 8>(blk, X0, X1, C, k, beta, stream);
+    case 16: return launch_spmm_vg<VEC, 16>(blk, X0, X1, C, k, beta, stream);
+    case 32: return launch_spmm_vg<VEC, 32>(blk, X0, X1, C, k, beta, stream);
+    default: return launch_spmm_vg<VEC, 64>(blk, X0, X1, C, k, beta, stream);
   }
 }
 
@@ -348,14 +373,21 @@ int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
   blk.n_items = (int64_t)item_row.size();
   blk.n_split_rows = (int64_t)split_rows.size();
 
+  // pack (col, val) into 8-byte pairs for the staged kernel loads
+  std::vector<int2> pairs((size_t)nnz);
+  for (int64_t t = 0; t < nnz; ++t) {
+    pairs[(size_t)t].x = indices[t];
+    float v = data[t];
+    pairs[(size_t)t].y = *reinterpret_cast<const int32_t *>(&v);
+  }
+
   auto upload = [&](void **dst, const void *src, size_t bytes) -> int {
     if (bytes == 0) { *dst = nullptr; return 0; }
     HIP_CHECK(hipMalloc(dst, bytes));
     HIP_CHECK(hipMemcpy(*dst, src, bytes, hipMemcpyHostToDevice));
     return 0;
   };
-  if (upload((void **)&blk.indices, indices, nnz * sizeof(int32_t)) ||
-      upload((void **)&blk.data, data, nnz * sizeof(float)) ||
+  if (upload((void **)&blk.pairs, pairs.data(), pairs.size() * sizeof(int2)) ||
       upload((void **)&blk.item_row, item_row.data(), item_row.size() * 4) ||
       upload((void **)&blk.item_begin, item_begin.data(), item_begin.size() * 4) ||
       upload((void **)&blk.item_end, item_end.data(), item_end.size() * 4) ||
@@ -374,7 +406,7 @@ int arrow_csr_destroy(int64_t handle) {
     return -1;
   }
   CsrBlock &b = it->second;
-  for (void *p : {(void *)b.indices, (void *)b.data, (void *)b.item_row,
+  for (void *p : {(void *)b.pairs, (void *)b.item_row,
                   (void *)b.item_begin, (void *)b.item_end,
                   (void *)b.split_rows}) {
     if (p) (void)hipFree(p);
@@ -389,14 +421,14 @@ int64_t arrow_csr_nnz(int64_t handle) {
   return it->second.nnz;
 }
 
-int arrow_spmm(int64_t handle, const float *X_dev, float *C_dev, int64_t k,
-               int beta, void *stream_v) {
+int arrow_spmm_dual(int64_t handle, const float *X0_dev, const float *X1_dev,
+                    float *C_dev, int64_t k, int beta, void *stream_v) {
   auto it = g_blocks.find(handle);
   if (it == g_blocks.end()) {
     set_error("arrow_spmm: bad handle");
     return -1;
   }
-  if (k <= 0 || !X_dev || !C_dev) {
+  if (k <= 0 || !X0_dev || !X1_dev || !C_dev) {
     set_error("arrow_spmm: bad arguments");
     return -1;
   }
@@ -414,10 +446,15 @@ int arrow_spmm(int64_t handle, const float *X_dev, float *C_dev, int64_t k,
 
   const LaunchCfg cfg = pick_cfg(k);
   switch (cfg.vec) {
-    case 4: return launch_spmm_v<4>(blk, X_dev, C_dev, k, beta, cfg.group, stream);
-    case 2: return launch_spmm_v<2>(blk, X_dev, C_dev, k, beta, cfg.group, stream);
-    default: return launch_spmm_v<1>(blk, X_dev, C_dev, k, beta, cfg.group, stream);
+    case 4: return launch_spmm_v<4>(blk, X0_dev, X1_dev, C_dev, k, beta, cfg.group, stream);
+    case 2: return launch_spmm_v<2>(blk, X0_dev, X1_dev, C_dev, k, beta, cfg.group, stream);
+    default: return launch_spmm_v<1>(blk, X0_dev, X1_dev, C_dev, k, beta, cfg.group, stream);
   }
+}
+
+int arrow_spmm(int64_t handle, const float *X_dev, float *C_dev, int64_t k,
+               int beta, void *stream_v) {
+  return arrow_spmm_dual(handle, X_dev, X_dev, C_dev, k, beta, stream_v);
 }
 
 int arrow_gather_rows_f32(const float *src, float *dst, const int64_t *idx,

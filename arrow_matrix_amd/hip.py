@@ -47,6 +47,10 @@ def _load():
     lib.arrow_spmm.argtypes = [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
                                ctypes.c_int64, ctypes.c_int, ctypes.c_void_p]
     lib.arrow_spmm.restype = ctypes.c_int
+    lib.arrow_spmm_dual.argtypes = [ctypes.c_int64, ctypes.c_void_p,
+                                    ctypes.c_void_p, ctypes.c_void_p,
+                                    ctypes.c_int64, ctypes.c_int, ctypes.c_void_p]
+    lib.arrow_spmm_dual.restype = ctypes.c_int
     for name in ('arrow_gather_rows_f32', 'arrow_scatter_rows_f32',
                  'arrow_scatter_add_rows_f32'):
         fn = getattr(lib, name)
@@ -71,15 +75,22 @@ class CsrBlockGPU:
     """A CSR block resident in HBM (uploaded ONCE — unlike the reference,
     which re-uploads A every iteration, arrow_slim_mpi.py:184-232)."""
 
-    def __init__(self, csr):
+    def __init__(self, csr=None, arrays=None):
+        """csr: scipy CSR, or arrays=(shape, indptr, indices, data) for raw
+        uploads (the fused layouts use negative column indices, which scipy
+        would reject)."""
         lib = _load()
-        csr = csr.tocsr()
-        rows, cols = csr.shape
-        indptr = np.ascontiguousarray(csr.indptr, dtype=np.int64)
-        indices = np.ascontiguousarray(csr.indices, dtype=np.int32)
-        data = np.ascontiguousarray(csr.data, dtype=np.float32)
+        if arrays is not None:
+            (rows, cols), indptr, indices, data = arrays
+        else:
+            csr = csr.tocsr()
+            rows, cols = csr.shape
+            indptr, indices, data = csr.indptr, csr.indices, csr.data
+        indptr = np.ascontiguousarray(indptr, dtype=np.int64)
+        indices = np.ascontiguousarray(indices, dtype=np.int32)
+        data = np.ascontiguousarray(data, dtype=np.float32)
         self.shape = (rows, cols)
-        self.nnz = int(csr.nnz)
+        self.nnz = int(indices.size)
         self._handle = _check(lib.arrow_csr_create(
             rows, cols, self.nnz,
             indptr.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
@@ -91,6 +102,13 @@ class CsrBlockGPU:
         """C (+)= A @ X on device pointers (e.g. torch tensor data_ptr())."""
         _check(_load().arrow_spmm(self._handle, X_ptr, C_ptr, k, beta, stream),
                "arrow_spmm")
+
+    def spmm_dual(self, X0_ptr: int, X1_ptr: int, C_ptr: int, k: int,
+                  beta: int, stream: int = 0):
+        """Fused dual-operand SpMM: negative column indices (encoded at
+        upload as -(idx+1)) read X1 (see include/arrow_spmm.h)."""
+        _check(_load().arrow_spmm_dual(self._handle, X0_ptr, X1_ptr, C_ptr,
+                                       k, beta, stream), "arrow_spmm_dual")
 
     def __del__(self):
         if getattr(self, '_handle', None) is not None and _lib is not None:

@@ -273,9 +273,11 @@ def main():
     if use_gpu and events:
         total_ms = 0.0
         total_bytes = 0.0
-        for s, e, nnz, rows in events:
+        for s, e, nnz, c_rows, x_rows in events:
             total_ms += s.elapsed_time(e)
-            total_bytes += 8.0 * nnz + 4.0 * (rows + 1) + 8.0 * k * rows
+            # algorithmic bytes: A pairs once + work items +
+            # X rows once + C rows written once
+            total_bytes += 8.0 * nnz + 4.0 * (c_rows + 1) + 4.0 * k * (x_rows + c_rows)
         achieved = total_bytes / (total_ms / 1e3) / 1e9  # GB/s
         peak = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
         roofline = {

@@ -61,6 +61,14 @@ int64_t arrow_csr_nnz(int64_t handle);
 int arrow_spmm(int64_t handle, const float *X_dev, float *C_dev, int64_t k,
                int beta, void *stream);
 
+/* Dual-operand SpMM for the FUSED diagonal + first-block-column layout
+ * (replaces the reference's back-to-back C_i = A_ii@X_i; C_i += A_i0@X_0,
+ * arrow_slim_mpi.py:121-144, writing C once instead of read-modify-write):
+ * a column index c >= 0 reads X0[c]; c < 0 reads X1[-c-1]. The negative
+ * encoding is fixed at arrow_csr_create time by the caller's indices. */
+int arrow_spmm_dual(int64_t handle, const float *X0_dev, const float *X1_dev,
+                    float *C_dev, int64_t k, int beta, void *stream);
+
 /* dst[i, :] = src[idx[i], :]   (n rows of width k, fp32, device) */
 int arrow_gather_rows_f32(const float *src_dev, float *dst_dev,
                           const int64_t *idx_dev, int64_t n, int64_t k,

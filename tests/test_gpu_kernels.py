@@ -93,6 +93,39 @@ def test_spmm_single_row_single_col(gpu):
     _check_spmm(gpu, A, 7, beta=0)
 
 
+def test_spmm_dual_negative_columns(gpu):
+    """arrow_spmm_dual: negative-encoded columns read the second operand."""
+    rng = np.random.default_rng(7)
+    rows, n0, n1, k = 200, 150, 60, 32
+    A0 = _random_csr(rows, n0, 0.05, seed=70)
+    A1 = _random_csr(rows, n1, 0.08, seed=71)
+    # merged arrays: per row, A0 entries then A1 entries (negative encoding)
+    indptr = np.zeros(rows + 1, dtype=np.int64)
+    cols, vals = [], []
+    for r in range(rows):
+        c0 = A0.indices[A0.indptr[r]:A0.indptr[r + 1]]
+        v0 = A0.data[A0.indptr[r]:A0.indptr[r + 1]]
+        c1 = A1.indices[A1.indptr[r]:A1.indptr[r + 1]]
+        v1 = A1.data[A1.indptr[r]:A1.indptr[r + 1]]
+        cols.append(np.concatenate([c0.astype(np.int64),
+                                    -(c1.astype(np.int64) + 1)]))
+        vals.append(np.concatenate([v0, v1]))
+        indptr[r + 1] = indptr[r] + cols[-1].size
+    cols = np.concatenate(cols).astype(np.int32)
+    vals = np.concatenate(vals).astype(np.float32)
+    blk = gpu.CsrBlockGPU(arrays=((rows, n0), indptr, cols, vals))
+    X0 = (2 * rng.random((n0, k)) - 1).astype(np.float32)
+    X1 = (2 * rng.random((n1, k)) - 1).astype(np.float32)
+    X0t = torch.from_numpy(X0).cuda()
+    X1t = torch.from_numpy(X1).cuda()
+    Ct = torch.empty((rows, k), dtype=torch.float32, device='cuda')
+    blk.spmm_dual(X0t.data_ptr(), X1t.data_ptr(), Ct.data_ptr(), k, 0,
+                  torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    ref = A0 @ X0 + A1 @ X1
+    np.testing.assert_allclose(Ct.cpu().numpy(), ref, rtol=1e-5, atol=1e-5)
+
+
 def test_gather_scatter_roundtrip(gpu):
     rng = np.random.default_rng(0)
     for k in (4, 16, 33, 128):
