@@ -231,14 +231,19 @@ class ArrowSlimMPI(ArrowMatrix):
         self._select_result_buffer()
         w = self.width
 
-        # X_0 broadcast (arrow_slim_mpi.py:265-273); owner of block 0 is rank 0
+        # X_0 broadcast (arrow_slim_mpi.py:265-273); owner of block 0 is
+        # rank 0. Overlapped: the row-0 launch below reads only the stripe,
+        # so the broadcast runs concurrently with it (the reference's
+        # broadcast is blocking, arrow_slim_mpi.py:273).
         if self.first_block == 0 and self.n_owned > 0:
             self.X_0.copy_(self.X_i[:w])
-        self.comm.bcast_(self.X_0, src=0)
+        bcast_work = self.comm.bcast_(self.X_0, src=0, async_op=True)
 
         if self.backend.device == 'cuda':
-            self._spmm_gpu()
+            self._spmm_gpu(bcast_work)
         else:
+            if bcast_work is not None:
+                bcast_work.wait()
             self._spmm_cpu()
 
         # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155)
@@ -257,9 +262,11 @@ class ArrowSlimMPI(ArrowMatrix):
         e.record()
         self.kernel_events.append((s, e, nnz, c_rows, x_rows))
 
-    def _spmm_gpu(self) -> None:
+    def _spmm_gpu(self, bcast_work=None) -> None:
         """Two fused launches on the resident merged structures
-        (DESIGN.md §kernels)."""
+        (DESIGN.md §kernels), with the X_0 broadcast and the C_0 reduce
+        overlapped with compute (RCCL runs on its own stream; `wait()`
+        inserts stream dependencies, not host blocks)."""
         be = self.backend
         w = self.width
 
@@ -272,18 +279,23 @@ class ArrowSlimMPI(ArrowMatrix):
         else:
             self.C_0.zero_()
 
-        # ONE reduce of the first block-row partials (arrow_slim_mpi.py:116)
-        self.comm.reduce_sum_(self.C_0, dst=0)
+        # ONE reduce of the first block-row partials (arrow_slim_mpi.py:116),
+        # overlapped with the rest launch below
+        reduce_work = self.comm.reduce_sum_(self.C_0, dst=0, async_op=True)
 
         # C_rest = A_diag_merged @ X_stripe + A_col_merged @ X_0 fused:
         # C written once (vs the reference's multiply-then-accumulate,
         # arrow_slim_mpi.py:121-144)
+        if bcast_work is not None:
+            bcast_work.wait()  # rest reads X_0
         if self._A_rest is not None:
             h = self._A_rest
             C_sub = self.C_i[self._rest_row_offset:
                              self.n_owned * w]
             self._timed(lambda: be.spmm_dual(h, self.X_i, self.X_0, C_sub, 0),
                         h.nnz, C_sub.shape[0], h.x_rows)
+        if reduce_work is not None:
+            reduce_work.wait()
 
     def _spmm_cpu(self) -> None:
         """Per-block scipy dataflow — the reference's own cpu path

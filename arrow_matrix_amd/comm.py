@@ -30,11 +30,11 @@ class Comm:
     def barrier(self):
         pass
 
-    def bcast_(self, tensor: torch.Tensor, src: int):
-        pass
+    def bcast_(self, tensor: torch.Tensor, src: int, async_op: bool = False):
+        return None
 
-    def reduce_sum_(self, tensor: torch.Tensor, dst: int):
-        pass
+    def reduce_sum_(self, tensor: torch.Tensor, dst: int, async_op: bool = False):
+        return None
 
     def allreduce_max_(self, tensor: torch.Tensor):
         pass
@@ -70,11 +70,13 @@ class TorchDistComm(Comm):
     def barrier(self):
         dist.barrier(group=self.group)
 
-    def bcast_(self, tensor: torch.Tensor, src: int):
-        dist.broadcast(tensor, src=src, group=self.group)
+    def bcast_(self, tensor: torch.Tensor, src: int, async_op: bool = False):
+        return dist.broadcast(tensor, src=src, group=self.group,
+                              async_op=async_op)
 
-    def reduce_sum_(self, tensor: torch.Tensor, dst: int):
-        dist.reduce(tensor, dst=dst, op=dist.ReduceOp.SUM, group=self.group)
+    def reduce_sum_(self, tensor: torch.Tensor, dst: int, async_op: bool = False):
+        return dist.reduce(tensor, dst=dst, op=dist.ReduceOp.SUM,
+                           group=self.group, async_op=async_op)
 
     def allreduce_max_(self, tensor: torch.Tensor):
         dist.all_reduce(tensor, op=dist.ReduceOp.MAX, group=self.group)
