@@ -89,7 +89,17 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
   const int64_t col0 = col_off + (int64_t)lane_in_group * VEC;
   const bool active = !GUARD || (col0 < k);
 
-  int64_t item = (int64_t)blockIdx.x * GROUPS_PER_BLOCK + group_in_block;
+  // XCD-aware workgroup remap (performance only): the dispatcher places
+  // block b on XCD b%8, so remap block ids to give each XCD a CONTIGUOUS
+  // range of work items — consecutive rows of a banded block then share the
+  // XCD's private L2 window instead of interleaving across all 8 L2s.
+  // Bijective form (cdna_hip_programming.md §XCD swizzle).
+  const int nwg = gridDim.x;
+  const int q = nwg / 8, rm = nwg % 8;
+  const int xcd = blockIdx.x % 8, pos = blockIdx.x / 8;
+  const int wg = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + pos;
+
+  int64_t item = (int64_t)wg * GROUPS_PER_BLOCK + group_in_block;
   const int64_t stride = (int64_t)gridDim.x * GROUPS_PER_BLOCK;
 
   for (; item < n_items; item += stride) {
