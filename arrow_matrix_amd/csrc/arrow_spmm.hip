@@ -27,6 +27,7 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
+#include <cstdlib>
 #include <string>
 #include <unordered_map>
 #include <vector>
@@ -82,7 +83,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
     const int32_t *__restrict__ item_row, const int32_t *__restrict__ item_begin,
     const int32_t *__restrict__ item_end, int64_t n_items,
     const float *__restrict__ X0, const float *__restrict__ X1,
-    float *__restrict__ C, int64_t k, int64_t col_off) {
+    float *__restrict__ C, int64_t k, int64_t col_off, int xcd_remap) {
   constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
   const int lane_in_group = threadIdx.x % GROUP;
   const int group_in_block = threadIdx.x / GROUP;
@@ -94,10 +95,13 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
   // range of work items — consecutive rows of a banded block then share the
   // XCD's private L2 window instead of interleaving across all 8 L2s.
   // Bijective form (cdna_hip_programming.md §XCD swizzle).
-  const int nwg = gridDim.x;
-  const int q = nwg / 8, rm = nwg % 8;
-  const int xcd = blockIdx.x % 8, pos = blockIdx.x / 8;
-  const int wg = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + pos;
+  int wg = blockIdx.x;
+  if (xcd_remap) {
+    const int nwg = gridDim.x;
+    const int q = nwg / 8, rm = nwg % 8;
+    const int xcd = blockIdx.x % 8, pos = blockIdx.x / 8;
+    wg = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + pos;
+  }
 
   int64_t item = (int64_t)wg * GROUPS_PER_BLOCK + group_in_block;
   const int64_t stride = (int64_t)gridDim.x * GROUPS_PER_BLOCK;
@@ -287,9 +291,13 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
   for (int64_t col_off = 0; col_off < k; col_off += span) {
     const bool guard = (col_off + span > k);
     auto run = [&](auto kern) {
+      static const int xcd_remap =
+          [] { const char *e = getenv("ARROW_SPMM_NO_XCD_REMAP");
+               return (e && e[0] == '1') ? 0 : 1; }();
       hipLaunchKernelGGL(kern, dim3(blocks), dim3(BLOCK_THREADS), 0, stream,
                          blk.pairs, blk.item_row, blk.item_begin,
-                         blk.item_end, blk.n_items, X0, X1, C, k, col_off);
+                         blk.item_end, blk.n_items, X0, X1, C, k, col_off,
+                         xcd_remap);
     };
     if (beta == 0) {
       if (guard) run(spmm_kernel<VEC, GROUP, 0, true>);
