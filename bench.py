@@ -52,6 +52,9 @@ def build_parser():
                    help='diagonal-block band half-width (planar-like locality)')
     p.add_argument('--no-cpu-baseline', action='store_true')
     p.add_argument('--json-out', type=str, default=None)
+    p.add_argument('--graph', choices=['auto', 'on', 'off'], default='auto',
+                   help='hipGraph-capture the steady-state iteration '
+                        '(BASELINE cfg5); auto = on for single-GPU runs')
     return p
 
 
@@ -241,24 +244,50 @@ def main():
         arrow.step()
         eng0.set_features(eng0.result_tile())
 
+    use_graph = use_gpu and (args.graph == 'on'
+                             or (args.graph == 'auto' and world <= 1))
+    steps = args.steps
+    if use_graph and steps % 2 == 1:
+        steps += 1  # the ping-pong X<->C period is 2 steps (see below)
+
     for _ in range(args.warmup):
         one_step()
     sync()
 
-    # instrument the SpMM kernel launches for the roofline
+    # roofline instrumentation pass (HIP events on the launch stream; kept
+    # OUTSIDE the timed region so the timed loop is uninstrumented)
     events = []
     for eng in arrow.engines:
         eng.kernel_events = events
+    for _ in range(2):
+        one_step()
+    sync()
+    for eng in arrow.engines:
+        eng.kernel_events = None
+
+    graph = None
+    if use_graph:
+        # hipGraph-captured steady-state iteration (BASELINE cfg5). One
+        # capture spans TWO steps: the X<->C ping-pong returns to its
+        # original buffer assignment after a pair, so a replay is exactly
+        # 2 iterations.
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            one_step()
+            one_step()
+        graph.replay()
+        sync()
 
     sync()
     t_start = time.perf_counter()
-    for _ in range(args.steps):
-        one_step()
+    if graph is not None:
+        for _ in range(steps // 2):
+            graph.replay()
+    else:
+        for _ in range(steps):
+            one_step()
     sync()
     elapsed = time.perf_counter() - t_start
-
-    for eng in arrow.engines:
-        eng.kernel_events = None
 
     # MAX over ranks
     t_t = torch.tensor([elapsed], dtype=torch.float64,
@@ -267,7 +296,7 @@ def main():
     elapsed = float(t_t.item())
 
     nnz_total = expected_nnz(w, nb) * L
-    flops = 2.0 * nnz_total * k * args.steps
+    flops = 2.0 * nnz_total * k * steps
     gflops = flops / elapsed / 1e9
 
     # roofline from the kernel events (this rank; rank 0 reports)
@@ -304,9 +333,9 @@ def main():
             "value": round(gflops, 2),
             "unit": "GFLOP/s",
             "n_gpus": n_gpus,
-            "steps": args.steps,
+            "steps": steps,
             "warmup": args.warmup,
-            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "ms_per_step": round(elapsed / steps * 1e3, 3),
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
@@ -317,6 +346,7 @@ def main():
                 "rows": args.rows, "width": w, "n_blocks": nb, "parts": L,
                 "features": k, "nnz": nnz_total, "band": args.band,
                 "parallelism": f"block-rows over {n_gpus} GPU(s), RCCL/xGMI",
+                "hipgraph": bool(use_graph),
             },
             "roofline": roofline,
             "cpu_baseline": cpu_base,
