@@ -32,10 +32,7 @@ def bench_spmm(path: Union[str, None],
                slim: bool = True,
                npy_format: bool = True):
     assert width > 0
-    if not slim or not blocked:
-        raise NotImplementedError(
-            "the ArrowMPI split/banded path (--slim False / --blocked False) "
-            "is not built yet — see DESIGN.md §next (SURVEY.md §8f-1)")
+    assert not slim or blocked  # reference arrow_dec_mpi.py:131
     if not npy_format:
         raise NotImplementedError(
             "only the .npy CSR on-disk format is supported (--npy True)")

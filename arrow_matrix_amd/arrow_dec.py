@@ -81,12 +81,16 @@ class ArrowDecompositionMPI:
         (arrow_dec_mpi.py:106-177). to_prev/to_next are the per-rank
         permutation slices as returned by load_decomposition_new (lists with
         one entry per decomposition part)."""
-        assert slim and block_diagonal, \
-            "round 1 implements the slim block-diagonal path (reference default); " \
-            "ArrowMPI split/banded mode is tracked in DESIGN.md §next"
+        assert not slim or block_diagonal  # reference arrow_dec_mpi.py:131
         comm = comm if comm is not None else default_comm()
-        engines = [ArrowSlimMPI(comm, tiles_per_side=int(nb), device=device)
-                   for nb in n_blocks]
+        if slim:
+            engines = [ArrowSlimMPI(comm, tiles_per_side=int(nb), device=device)
+                       for nb in n_blocks]
+        else:
+            from .arrow_mpi import ArrowMPI
+            engines = [ArrowMPI(comm, is_block_diagonal=block_diagonal,
+                                tiles_per_side=int(nb), device=device)
+                       for nb in n_blocks]
         return ArrowDecompositionMPI(comm, engines, n_blocks, rows_per_rank,
                                      feature_columns, to_prev_permutation,
                                      to_next_permutation, device=device)
@@ -265,6 +269,13 @@ class ArrowDecompositionMPI:
             for r in range(max(first, 1), last):
                 grid[r][r] = _extract_block(A, r, r, width)
                 grid[r][0] = _extract_block(A, r, 0, width)
+                if not is_block_diagonal:
+                    # banded off-diagonals (reference sends A_ii-1/A_ii+1,
+                    # arrow_dec_mpi.py:804-815)
+                    if r >= 2:
+                        grid[r][r - 1] = _extract_block(A, r, r - 1, width)
+                    if r < nb - 1:
+                        grid[r][r + 1] = _extract_block(A, r, r + 1, width)
             blocks_per_part.append(grid)
             w = width
             tp = to_prev_full[i]

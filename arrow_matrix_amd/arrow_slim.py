@@ -35,11 +35,15 @@ from .comm import Comm
 class ArrowSlimMPI(ArrowMatrix):
 
     def __init__(self, comm: Optional[Comm] = None, tiles_per_side: Optional[int] = None,
-                 device: str = 'cpu'):
+                 device: str = 'cpu', banded: bool = False):
         """:param comm: communicator over this matrix's ranks (None = single
         process). :param tiles_per_side: number of block-rows (n_blocks);
-        defaults to comm.size, the reference's one-rank-per-block layout."""
+        defaults to comm.size, the reference's one-rank-per-block layout.
+        :param banded: arrow-banded decomposition (blocks (r, r±1) carried;
+        ±1 halo X exchange per iteration — the ArrowMPI variant,
+        arrow_mpi.py:123-175,211-219)."""
         self.comm = comm if comm is not None else Comm()
+        self.banded = banded
         self.column_comm = self.comm
         self.tiles_per_side = tiles_per_side if tiles_per_side is not None else self.comm.size
         assert self.tiles_per_side >= 1
@@ -62,6 +66,12 @@ class ArrowSlimMPI(ArrowMatrix):
         self.C_i: Optional[torch.Tensor] = None   # (n_owned*width, k) stripe
         self.C_0: Optional[torch.Tensor] = None   # (width, k) partial/reduced
         self.nnz_owned = 0
+        self.A_lo: List = []          # A_{r,r-1} for owned r >= 2 (banded)
+        self.A_hi: List = []          # A_{r,r+1} for owned r < nb-1 (banded)
+        self.X_halo_lo = None         # X_{first-1} tile from the prev rank
+        self.X_halo_hi = None         # X_{last} tile from the next rank
+        self._A_bd_lo = None          # boundary (first, first-1) vs halo_lo
+        self._A_bd_hi = None          # boundary (last-1, last) vs halo_hi
         # merged resident GPU structures (built at load time, gpu only)
         self._A_row0 = None
         self._A_rest = None
@@ -89,7 +99,9 @@ class ArrowSlimMPI(ArrowMatrix):
         CPU path keeps per-block scipy (the reference's own cpu dataflow).
         """
         assert len(blocks) == self.tiles_per_side
+        nb = self.tiles_per_side
         self.A_0i, self.A_ii, self.A_i0 = [], [], []
+        self.A_lo, self.A_hi = [], []
         self.nnz_owned = 0
         gpu = self.backend.device == 'cuda'
         for r in range(self.first_block, self.last_block):
@@ -109,10 +121,19 @@ class ArrowSlimMPI(ArrowMatrix):
             else:
                 self.A_ii.append(None)
                 self.A_i0.append(None)
+            # banded off-diagonals (arrow_mpi.py:459-469): (r, r-1) for
+            # r >= 2, (r, r+1) for 0 < r < nb-1
+            lo = blocks[r][r - 1] if (self.banded and r >= 2) else None
+            hi = blocks[r][r + 1] if (self.banded and 0 < r < nb - 1) else None
+            self.A_lo.append(lo if (gpu or lo is None) else self.backend.upload_block(lo))
+            self.A_hi.append(hi if (gpu or hi is None) else self.backend.upload_block(hi))
+            self.nnz_owned += (lo.nnz if lo is not None else 0) + \
+                              (hi.nnz if hi is not None else 0)
         if gpu:
             self._build_merged_gpu()
             # drop the host block references — device copies are resident
             self.A_0i = self.A_ii = self.A_i0 = None
+            self.A_lo = self.A_hi = None
 
     def _build_merged_gpu(self) -> None:
         w = self.width
@@ -131,8 +152,14 @@ class ArrowSlimMPI(ArrowMatrix):
             data_cat.append(b.data.astype(np.float32))
         self._A_row0 = self._merged_handle(w, nw * w, rows_cat, cols_cat,
                                            data_cat, x_rows=nw * w)
-        # --- rest merge: C[r] = A_rr @ X_r  +  A_r0 @ X_0 ------------------
+        # --- rest merge: C[r] = A_rr @ X_r + A_r0 @ X_0
+        #     (+ interior banded off-diagonals A_{r,r±1} @ X_{r±1} when the
+        #      neighbour block is owned; boundary off-diagonals become small
+        #      separate structures against the halo tiles) ----------------
         self._rest_row_offset = w if self.first_block == 0 else 0
+        self._A_bd_lo = self._A_bd_hi = None
+        lo_list = self.A_lo if self.A_lo else [None] * nw
+        hi_list = self.A_hi if self.A_hi else [None] * nw
         rows_cat, cols_cat, data_cat = [], [], []
         n_rest = 0
         for j, r in enumerate(range(self.first_block, self.last_block)):
@@ -147,6 +174,19 @@ class ArrowSlimMPI(ArrowMatrix):
             rows_cat.append(local + np.repeat(np.arange(w), np.diff(r0.indptr)))
             cols_cat.append(-(r0.indices.astype(np.int64) + 1))  # -> X_0
             data_cat.append(r0.data.astype(np.float32))
+            for off, blk in ((-1, lo_list[j]), (1, hi_list[j])):
+                if blk is None:
+                    continue
+                t = r + off
+                if self.first_block <= t < self.last_block:  # interior
+                    bb = blk.tocsr()
+                    rows_cat.append(local + np.repeat(np.arange(w), np.diff(bb.indptr)))
+                    cols_cat.append(bb.indices.astype(np.int64) + (t - self.first_block) * w)
+                    data_cat.append(bb.data.astype(np.float32))
+                elif off < 0:  # boundary vs X_halo_lo
+                    self._A_bd_lo = (self.backend.upload_block(blk), local)
+                else:          # boundary vs X_halo_hi
+                    self._A_bd_hi = (self.backend.upload_block(blk), local)
             n_rest += 1
         if n_rest:
             rest_rows = nw * w - self._rest_row_offset
@@ -184,6 +224,13 @@ class ArrowSlimMPI(ArrowMatrix):
                 setattr(self, name, self.backend.zeros(shape))
             else:
                 buf.zero_()
+        if self.banded:
+            for name in ('X_halo_lo', 'X_halo_hi'):
+                buf = getattr(self, name)
+                if buf is None or tuple(buf.shape) != (w, k):
+                    setattr(self, name, self.backend.zeros((w, k)))
+                else:
+                    buf.zero_()
         # ping-pong pool: spmm writes C into a stripe that does NOT alias
         # X_i, so `X := C` between iterations (set_features(result_tile()))
         # is race-free on the GPU (the reference allocates a fresh C every
@@ -238,6 +285,9 @@ class ArrowSlimMPI(ArrowMatrix):
         if self.first_block == 0 and self.n_owned > 0:
             self.X_0.copy_(self.X_i[:w])
         bcast_work = self.comm.bcast_(self.X_0, src=0, async_op=True)
+
+        if self.banded:
+            self._exchange_halos()
 
         if self.backend.device == 'cuda':
             self._spmm_gpu(bcast_work)
@@ -294,8 +344,46 @@ class ArrowSlimMPI(ArrowMatrix):
                              self.n_owned * w]
             self._timed(lambda: be.spmm_dual(h, self.X_i, self.X_0, C_sub, 0),
                         h.nnz, C_sub.shape[0], h.x_rows)
+        # banded boundary off-diagonals against the received halo tiles
+        for bd, halo in ((self._A_bd_lo, self.X_halo_lo),
+                         (self._A_bd_hi, self.X_halo_hi)):
+            if bd is not None:
+                hdl, local = bd
+                Cr = self.C_i[local:local + w]
+                self._timed(lambda: be.spmm_block(hdl, halo, Cr, 1),
+                            hdl.nnz, w, w)
         if reduce_work is not None:
             reduce_work.wait()
+
+    def _exchange_halos(self) -> None:
+        """±1 halo X exchange for the banded mode: my first owned tile goes
+        to the previous rank (their halo_hi), my last to the next rank
+        (their halo_lo) — replaces the reference's Isend/Irecv halo pattern
+        (arrow_mpi.py:123-175) with grouped p2p over xGMI."""
+        if self.comm.size == 1 or self.n_owned == 0:
+            return
+        import torch.distributed as dist
+        w = self.width
+        P = self.comm.size
+        bpr = self.blocks_per_rank
+        prev_rank = (self.first_block - 1) // bpr if self.first_block >= 1 else -1
+        next_owner = self.last_block // bpr if self.last_block < self.tiles_per_side else -1
+        ops = []
+        group = getattr(self.comm, 'group', None)
+        if prev_rank >= 0:
+            ops.append(dist.P2POp(dist.isend, self.X_i[:w].contiguous(),
+                                  prev_rank, group=group))
+            ops.append(dist.P2POp(dist.irecv, self.X_halo_lo, prev_rank,
+                                  group=group))
+        if next_owner >= 0 and next_owner < P:
+            ops.append(dist.P2POp(dist.isend,
+                                  self.X_i[(self.n_owned - 1) * w:self.n_owned * w].contiguous(),
+                                  next_owner, group=group))
+            ops.append(dist.P2POp(dist.irecv, self.X_halo_hi, next_owner,
+                                  group=group))
+        if ops:
+            for wk in dist.batch_isend_irecv(ops):
+                wk.wait()
 
     def _spmm_cpu(self) -> None:
         """Per-block scipy dataflow — the reference's own cpu path
@@ -320,6 +408,19 @@ class ArrowSlimMPI(ArrowMatrix):
             Cr = self.C_i[j * w:(j + 1) * w]
             be.spmm_block(self.A_ii[j], Xr.contiguous(), Cr, 0)
             be.spmm_block(self.A_i0[j], self.X_0, Cr, 1)
+            # banded ±1 halo terms (arrow_mpi.py:211-219)
+            lo = self.A_lo[j] if self.A_lo else None
+            hi = self.A_hi[j] if self.A_hi else None
+            for off, blk in ((-1, lo), (1, hi)):
+                if blk is None:
+                    continue
+                t = r + off
+                if self.first_block <= t < self.last_block:
+                    Xn = self.X_i[(t - self.first_block) * w:
+                                  (t - self.first_block + 1) * w].contiguous()
+                else:
+                    Xn = self.X_halo_lo if off < 0 else self.X_halo_hi
+                be.spmm_block(blk, Xn, Cr, 1)
 
     # -- result --------------------------------------------------------------
 

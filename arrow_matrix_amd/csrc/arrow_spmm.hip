@@ -120,13 +120,17 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
     // Stage GROUP (col,val) pairs cooperatively (ONE coalesced 8-B/lane
     // load per GROUP nonzeros), then broadcast them across the group with
     // wave shuffles — the A stream costs 1/GROUP memory instructions per
-    // nonzero instead of 2, and the X-row loads below can be issued back
-    // to back by the unrolled loop.
-    auto consume = [&](int2 mine, int u) {
+    // nonzero instead of 2.
+    auto xaddr = [&](int2 mine, int u) {
       const int32_t c = __shfl(mine.x, u, GROUP);
-      const float v = __int_as_float(__shfl(mine.y, u, GROUP));
-      const float *xr = (c < 0 ? X1 + (int64_t)(-c - 1) * k
-                               : X0 + (int64_t)c * k) + col0;
+      return (c < 0 ? X1 + (int64_t)(-c - 1) * k : X0 + (int64_t)c * k) + col0;
+    };
+    auto val = [&](int2 mine, int u) {
+      return __int_as_float(__shfl(mine.y, u, GROUP));
+    };
+    auto consume = [&](int2 mine, int u) {
+      const float *xr = xaddr(mine, u);
+      const float v = val(mine, u);
       if (active) {
         if constexpr (VEC == 4) {
           const float4 xv = *reinterpret_cast<const float4 *>(xr);
@@ -143,19 +147,50 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
         }
       }
     };
-    // full GROUP-sized chunks: constant trip count, partially unrolled so
-    // several X-row loads are in flight per group
+    // 4 entries at a time with the four X-row loads issued back to back
+    // BEFORE any use: the rolled one-at-a-time loop stalls on each load's
+    // s_waitcnt before the next can issue (in-order issue), which makes
+    // short power-law rows latency-bound.
+    auto consume4 = [&](int2 mine, int u) {
+      if constexpr (VEC == 4) {
+        const float *p0 = xaddr(mine, u + 0);
+        const float *p1 = xaddr(mine, u + 1);
+        const float *p2 = xaddr(mine, u + 2);
+        const float *p3 = xaddr(mine, u + 3);
+        const float v0 = val(mine, u + 0), v1 = val(mine, u + 1);
+        const float v2 = val(mine, u + 2), v3 = val(mine, u + 3);
+        if (active) {
+          const float4 a0 = *reinterpret_cast<const float4 *>(p0);
+          const float4 a1 = *reinterpret_cast<const float4 *>(p1);
+          const float4 a2 = *reinterpret_cast<const float4 *>(p2);
+          const float4 a3 = *reinterpret_cast<const float4 *>(p3);
+          acc[0] = fmaf(v0, a0.x, acc[0]); acc[1] = fmaf(v0, a0.y, acc[1]);
+          acc[2] = fmaf(v0, a0.z, acc[2]); acc[3] = fmaf(v0, a0.w, acc[3]);
+          acc[0] = fmaf(v1, a1.x, acc[0]); acc[1] = fmaf(v1, a1.y, acc[1]);
+          acc[2] = fmaf(v1, a1.z, acc[2]); acc[3] = fmaf(v1, a1.w, acc[3]);
+          acc[0] = fmaf(v2, a2.x, acc[0]); acc[1] = fmaf(v2, a2.y, acc[1]);
+          acc[2] = fmaf(v2, a2.z, acc[2]); acc[3] = fmaf(v2, a2.w, acc[3]);
+          acc[0] = fmaf(v3, a3.x, acc[0]); acc[1] = fmaf(v3, a3.y, acc[1]);
+          acc[2] = fmaf(v3, a3.z, acc[2]); acc[3] = fmaf(v3, a3.w, acc[3]);
+        }
+      } else {
+        consume(mine, u); consume(mine, u + 1);
+        consume(mine, u + 2); consume(mine, u + 3);
+      }
+    };
     int32_t base = b;
-    for (; base + GROUP <= e; base += GROUP) {
+    for (; base + GROUP <= e; base += GROUP) {  // full chunks
       const int2 mine = pairs[base + lane_in_group];
-#pragma unroll 8
-      for (int u = 0; u < GROUP; ++u) consume(mine, u);
+#pragma unroll
+      for (int u = 0; u < GROUP; u += 4) consume4(mine, u);
     }
     if (base < e) {  // remainder (< GROUP entries)
       const int32_t t = base + lane_in_group;
       const int2 mine = (t < e) ? pairs[t] : int2{0, 0};
       const int cnt = e - base;
-      for (int u = 0; u < cnt; ++u) consume(mine, u);
+      int u = 0;
+      for (; u + 4 <= cnt; u += 4) consume4(mine, u);
+      for (; u < cnt; ++u) consume(mine, u);
     }
 
     if (active) {
@@ -291,9 +326,12 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
   for (int64_t col_off = 0; col_off < k; col_off += span) {
     const bool guard = (col_off + span > k);
     auto run = [&](auto kern) {
+      // default OFF: the contiguous-per-XCD remap measured 16% SLOWER on
+      // the cfg4-shaped workload (profiles/r01_*) — the shared moving
+      // window over L3 beats 8 disjoint per-XCD windows here
       static const int xcd_remap =
-          [] { const char *e = getenv("ARROW_SPMM_NO_XCD_REMAP");
-               return (e && e[0] == '1') ? 0 : 1; }();
+          [] { const char *e = getenv("ARROW_SPMM_XCD_REMAP");
+               return (e && e[0] == '1') ? 1 : 0; }();
       hipLaunchKernelGGL(kern, dim3(blocks), dim3(BLOCK_THREADS), 0, stream,
                          blk.pairs, blk.item_row, blk.item_begin,
                          blk.item_end, blk.n_items, X0, X1, C, k, col_off,

@@ -35,23 +35,29 @@ def _worker(rank, port, tmpdir, result_q):
         comm = TorchDistComm()
 
         cases = [
-            ([4], 6, 5, 0),       # 2 blocks per rank
-            ([2], 5, 3, 1),       # reference layout: 1 block per rank
-            ([3, 2], 4, 4, 2),    # L=2: forward/backward alltoallv
-            ([4, 3, 2], 4, 6, 3), # L=3 cascade
+            ([4], 6, 5, 0, False),       # 2 blocks per rank
+            ([2], 5, 3, 1, False),       # reference layout: 1 block per rank
+            ([3, 2], 4, 4, 2, False),    # L=2: forward/backward alltoallv
+            ([4, 3, 2], 4, 6, 3, False), # L=3 cascade
+            ([4], 6, 4, 7, True),        # banded: cross-rank halo exchange
+            ([5], 4, 3, 8, True),        # banded, uneven spans (3+2)
         ]
-        for n_blocks, width, k, seed in cases:
+        for n_blocks, width, k, seed, banded in cases:
             decomp = synth.synth_arrow_decomposition(width, n_blocks,
-                                                     avg_deg=5, seed=seed)
+                                                     avg_deg=5, seed=seed,
+                                                     block_diagonal=not banded)
             prefix = os.path.join(tmpdir, f'g{seed}')
             if rank == 0:
-                graphio.save_decomposition_new(decomp, prefix, width)
+                graphio.save_decomposition_new(decomp, prefix, width,
+                                               block_diagonal=not banded)
             dist.barrier()
 
             blocks, nb, to_prev, to_next = ArrowDecompositionMPI.load_decomposition_new(
-                comm, prefix, width, is_block_diagonal=True)
+                comm, prefix, width, is_block_diagonal=not banded)
             arrow = ArrowDecompositionMPI.initialize(comm, nb, to_prev, to_next,
-                                                     width, k, device='cpu')
+                                                     width, k, device='cpu',
+                                                     block_diagonal=not banded,
+                                                     slim=not banded)
             arrow.load_data_from_blocks(blocks)
             arrow.zero_rhs(width, k)
 
