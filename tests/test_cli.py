@@ -20,13 +20,17 @@ def test_cli_synthetic_cpu():
         assert 'FAILED' not in r.stdout
 
 
-def test_cli_slim_false_fails_clearly():
-    r = subprocess.run(
-        [sys.executable, os.path.join(REPO, 'scripts', 'spmm_arrow_main.py'),
-         '--width', '10', '--slim', 'false', '--device', 'cpu'],
-        capture_output=True, text=True, timeout=60)
-    assert r.returncode != 0
-    assert 'ArrowMPI' in (r.stderr + r.stdout)
+def test_cli_slim_false_runs_arrow_mpi():
+    """--slim False routes to the ArrowMPI engine (block-diagonal here)."""
+    with tempfile.TemporaryDirectory() as td:
+        r = subprocess.run(
+            [sys.executable, os.path.join(REPO, 'scripts', 'spmm_arrow_main.py'),
+             '--width', '12', '--features', '3', '--iterations', '1',
+             '--device', 'cpu', '--slim', 'false', '--ranksperside', '3',
+             '--ba_neighbors', '4'],
+            cwd=td, capture_output=True, text=True, timeout=120)
+        assert r.returncode == 0, r.stderr + r.stdout
+        assert 'FAILED' not in r.stdout
 
 
 def test_bench_spmm_importable_signature():
