@@ -33,10 +33,6 @@ def bench_spmm(path: Union[str, None],
                npy_format: bool = True):
     assert width > 0
     assert not slim or blocked  # reference arrow_dec_mpi.py:131
-    if not npy_format:
-        raise NotImplementedError(
-            "only the .npy CSR on-disk format is supported (--npy True)")
-
     comm = default_comm()
 
     if path is None:
@@ -44,8 +40,14 @@ def bench_spmm(path: Union[str, None],
         if comm.rank == 0:
             os.makedirs("tmp", exist_ok=True)
             decomp = synth.synth_arrow_decomposition(
-                width, [p_per_side], avg_deg=ba_neighbors, seed=503)
-            graphio.save_decomposition_new(decomp, path, width)
+                width, [p_per_side], avg_deg=ba_neighbors, seed=503,
+                block_diagonal=blocked)
+            if npy_format:
+                graphio.save_decomposition_new(decomp, path, width,
+                                               block_diagonal=blocked)
+            else:
+                graphio.save_decomposition(decomp, path, width,
+                                           block_diagonal=blocked)
             print("DATASET GENERATED -- ", p_per_side * width, " vertices", flush=True)
         comm.barrier()
 

@@ -219,16 +219,16 @@ class ArrowDecompositionMPI:
         one-based normalisation / overflow sentinel follow
         arrow_dec_mpi.py:680-749 exactly (tables.pad_and_compose_permutations).
         """
-        if not use_npy:
-            raise NotImplementedError(
-                "only the .npy CSR format is supported (use_npy=True); the "
-                "legacy .npz path is tracked in DESIGN.md §next")
         comm = comm if comm is not None else default_comm()
         P = comm.size
         rank = comm.rank
 
-        decomposition = graphio.load_decomposition_new(
-            filename, width, block_diagonal=is_block_diagonal, mem_map=use_mmap)
+        if use_npy:
+            decomposition = graphio.load_decomposition_new(
+                filename, width, block_diagonal=is_block_diagonal, mem_map=use_mmap)
+        else:
+            decomposition = graphio.load_decomposition(
+                filename, width, block_diagonal=is_block_diagonal)
         if len(decomposition) == 0:
             raise FileNotFoundError(
                 f"decomposition {filename!r} width={width} not found")

@@ -30,6 +30,8 @@ import torch
 from .arrow_matrix import ArrowMatrix
 from .backends import make_backend
 from .comm import Comm
+from .common import wb_logging
+import time
 
 
 class ArrowSlimMPI(ArrowMatrix):
@@ -282,19 +284,23 @@ class ArrowSlimMPI(ArrowMatrix):
         # rank 0. Overlapped: the row-0 launch below reads only the stripe,
         # so the broadcast runs concurrently with it (the reference's
         # broadcast is blocking, arrow_slim_mpi.py:273).
+        tic = time.perf_counter()
         if self.first_block == 0 and self.n_owned > 0:
             self.X_0.copy_(self.X_i[:w])
         bcast_work = self.comm.bcast_(self.X_0, src=0, async_op=True)
+        wb_logging.log({"spmm_x_bcast_time": time.perf_counter() - tic})
 
         if self.banded:
             self._exchange_halos()
 
+        tic = time.perf_counter()
         if self.backend.device == 'cuda':
             self._spmm_gpu(bcast_work)
         else:
             if bcast_work is not None:
                 bcast_work.wait()
             self._spmm_cpu()
+        wb_logging.log({"spmm_kernel_time": time.perf_counter() - tic})
 
         # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155)
         if self.first_block == 0 and self.n_owned > 0 and self.comm.rank == 0:
@@ -331,7 +337,9 @@ class ArrowSlimMPI(ArrowMatrix):
 
         # ONE reduce of the first block-row partials (arrow_slim_mpi.py:116),
         # overlapped with the rest launch below
+        tic = time.perf_counter()
         reduce_work = self.comm.reduce_sum_(self.C_0, dst=0, async_op=True)
+        wb_logging.log({"spmm_row_reduce": time.perf_counter() - tic})
 
         # C_rest = A_diag_merged @ X_stripe + A_col_merged @ X_0 fused:
         # C written once (vs the reference's multiply-then-accumulate,

@@ -122,6 +122,40 @@ def load_decomposition_new(filename: str, width: Optional[int] = None,
     return decomposition
 
 
+def load_decomposition(filename: str, width: Optional[int] = None,
+                       block_diagonal: bool = True, no_permutation: bool = False):
+    """Legacy .npz format (reference graphio.py:194-248): part i stored as
+    {prefix}_B_{width}_{i}[_bd].npz + ..._permutation.npy."""
+    decomposition = []
+    i = 0
+    while True:
+        path = format_path(filename, width, i, block_diagonal, DecompositionFileType.npz)
+        if not os.path.exists(path):
+            break
+        B = sparse.load_npz(path)
+        if no_permutation:
+            permutation = None
+        else:
+            p = format_path(filename, width, i, block_diagonal,
+                            DecompositionFileType.permutation_npy)
+            permutation = np.load(p)
+        decomposition.append((B, permutation))
+        i += 1
+    return decomposition
+
+
+def save_decomposition(decomposition, filename: str, width: int,
+                       block_diagonal: bool = True) -> None:
+    """Legacy .npz writer (reference graphio.py:103-117)."""
+    for i, (B, permutation) in enumerate(decomposition):
+        sparse.save_npz(format_path(filename, width, i, block_diagonal,
+                                    DecompositionFileType.npz),
+                        sparse.csr_matrix(B))
+        np.save(format_path(filename, width, i, block_diagonal,
+                            DecompositionFileType.permutation_npy),
+                np.asarray(permutation))
+
+
 def split_matrix_to_blocks(A: sparse.csr_matrix, block_size: int,
                            dtype=None, use_min_shape: bool = False
                            ) -> List[List[Union[sparse.csr_matrix, None]]]:

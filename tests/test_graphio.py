@@ -77,3 +77,19 @@ def test_split_matrix_to_blocks_vs_golden(golden):
                     np.testing.assert_array_equal(blocks[i][j].toarray(), golden[key])
                 else:
                     assert blocks[i][j] is None
+
+
+def test_legacy_npz_roundtrip_and_cli_path():
+    """Legacy .npz on-disk format (reference graphio.py:103-117,194-248)."""
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+    decomp = synth.synth_arrow_decomposition(4, [3], avg_deg=4, seed=21)
+    with tempfile.TemporaryDirectory() as td:
+        prefix = os.path.join(td, 'x')
+        graphio.save_decomposition(decomp, prefix, 4)
+        loaded = graphio.load_decomposition(prefix, 4)
+        assert len(loaded) == 1
+        diff = sparse.csr_matrix(decomp[0][0]) - sparse.csr_matrix(loaded[0][0])
+        assert diff.nnz == 0 or abs(diff).max() == 0
+        blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+            None, prefix, 4, use_npy=False)
+        assert int(nb[0]) == 3
