@@ -79,7 +79,14 @@ class TorchDistComm(Comm):
                            group=self.group, async_op=async_op)
 
     def allreduce_max_(self, tensor: torch.Tensor):
-        dist.all_reduce(tensor, op=dist.ReduceOp.MAX, group=self.group)
+        # RCCL needs device tensors; round-trip CPU flags (e.g. the
+        # fail-allreduce of arrow_bench) transparently
+        if tensor.device.type == 'cpu' and dist.get_backend(self.group) == 'nccl':
+            t = tensor.cuda()
+            dist.all_reduce(t, op=dist.ReduceOp.MAX, group=self.group)
+            tensor.copy_(t.cpu())
+        else:
+            dist.all_reduce(tensor, op=dist.ReduceOp.MAX, group=self.group)
 
     def alltoallv(self, send: torch.Tensor, send_counts: Sequence[int],
                   recv_counts: Sequence[int]) -> torch.Tensor:
