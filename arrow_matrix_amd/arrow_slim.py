@@ -74,6 +74,7 @@ class ArrowSlimMPI(ArrowMatrix):
         self.X_halo_hi = None         # X_{last} tile from the next rank
         self._A_bd_lo = None          # boundary (first, first-1) vs halo_lo
         self._A_bd_hi = None          # boundary (last-1, last) vs halo_hi
+        self._A_all = None            # single-process fully-fused structure
         # merged resident GPU structures (built at load time, gpu only)
         self._A_row0 = None
         self._A_rest = None
@@ -145,6 +146,8 @@ class ArrowSlimMPI(ArrowMatrix):
         self._rest_row_offset = 0
         if nw == 0:
             return
+        fuse_all = (self.comm.size == 1 and self.first_block == 0
+                    and self.n_owned == self.tiles_per_side)
         # --- row-0 merge: C_0 = [A_0,first .. A_0,last-1] @ X_stripe -------
         rows_cat, cols_cat, data_cat = [], [], []
         for j, b in enumerate(self.A_0i):
@@ -152,8 +155,15 @@ class ArrowSlimMPI(ArrowMatrix):
             rows_cat.append(np.repeat(np.arange(w), np.diff(b.indptr)))
             cols_cat.append(b.indices.astype(np.int64) + j * w)
             data_cat.append(b.data.astype(np.float32))
-        self._A_row0 = self._merged_handle(w, nw * w, rows_cat, cols_cat,
-                                           data_cat, x_rows=nw * w)
+        if fuse_all:
+            # single process: block-row 0's reduce is the identity, so the
+            # whole matrix fuses into ONE launch writing C in place — the
+            # X stripe is read once per iteration instead of twice, and
+            # X_0 == X_i[:w] needs no broadcast copy
+            row0_sets = (rows_cat, cols_cat, data_cat)
+        else:
+            self._A_row0 = self._merged_handle(w, nw * w, rows_cat, cols_cat,
+                                               data_cat, x_rows=nw * w)
         # --- rest merge: C[r] = A_rr @ X_r + A_r0 @ X_0
         #     (+ interior banded off-diagonals A_{r,r±1} @ X_{r±1} when the
         #      neighbour block is owned; boundary off-diagonals become small
@@ -162,7 +172,11 @@ class ArrowSlimMPI(ArrowMatrix):
         self._A_bd_lo = self._A_bd_hi = None
         lo_list = self.A_lo if self.A_lo else [None] * nw
         hi_list = self.A_hi if self.A_hi else [None] * nw
-        rows_cat, cols_cat, data_cat = [], [], []
+        if fuse_all:
+            self._rest_row_offset = 0
+            rows_cat, cols_cat, data_cat = row0_sets
+        else:
+            rows_cat, cols_cat, data_cat = [], [], []
         n_rest = 0
         for j, r in enumerate(range(self.first_block, self.last_block)):
             if r == 0:
@@ -190,7 +204,11 @@ class ArrowSlimMPI(ArrowMatrix):
                 else:          # boundary vs X_halo_hi
                     self._A_bd_hi = (self.backend.upload_block(blk), local)
             n_rest += 1
-        if n_rest:
+        if fuse_all:
+            self._A_all = self._merged_handle(nw * w, nw * w, rows_cat,
+                                              cols_cat, data_cat,
+                                              x_rows=nw * w + w)
+        elif n_rest:
             rest_rows = nw * w - self._rest_row_offset
             self._A_rest = self._merged_handle(rest_rows, nw * w, rows_cat,
                                                cols_cat, data_cat,
@@ -285,9 +303,11 @@ class ArrowSlimMPI(ArrowMatrix):
         # so the broadcast runs concurrently with it (the reference's
         # broadcast is blocking, arrow_slim_mpi.py:273).
         tic = time.perf_counter()
-        if self.first_block == 0 and self.n_owned > 0:
-            self.X_0.copy_(self.X_i[:w])
-        bcast_work = self.comm.bcast_(self.X_0, src=0, async_op=True)
+        bcast_work = None
+        if self._A_all is None:
+            if self.first_block == 0 and self.n_owned > 0:
+                self.X_0.copy_(self.X_i[:w])
+            bcast_work = self.comm.bcast_(self.X_0, src=0, async_op=True)
         wb_logging.log({"spmm_x_bcast_time": time.perf_counter() - tic})
 
         if self.banded:
@@ -302,8 +322,10 @@ class ArrowSlimMPI(ArrowMatrix):
             self._spmm_cpu()
         wb_logging.log({"spmm_kernel_time": time.perf_counter() - tic})
 
-        # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155)
-        if self.first_block == 0 and self.n_owned > 0 and self.comm.rank == 0:
+        # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155);
+        # in the fully-fused single-process layout it was written in place
+        if (self._A_all is None and self.first_block == 0
+                and self.n_owned > 0 and self.comm.rank == 0):
             self.C_i[:w].copy_(self.C_0)
 
     def _timed(self, fn, nnz, c_rows, x_rows):
@@ -325,6 +347,14 @@ class ArrowSlimMPI(ArrowMatrix):
         inserts stream dependencies, not host blocks)."""
         be = self.backend
         w = self.width
+
+        if self._A_all is not None:
+            # single process: whole matrix in ONE fused launch, C in place;
+            # X_0 == X_i[:w] so the second operand is the stripe itself
+            h = self._A_all
+            self._timed(lambda: be.spmm_dual(h, self.X_i, self.X_i, self.C_i, 0),
+                        h.nnz, self.n_owned * w, h.x_rows)
+            return
 
         # C_0 = A_row0_merged @ X_stripe  (ONE launch; the reference runs one
         # CSRMM per block and re-uploads A and X, arrow_slim_mpi.py:181-195)

@@ -181,8 +181,13 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
     int32_t base = b;
     for (; base + GROUP <= e; base += GROUP) {  // full chunks
       const int2 mine = pairs[base + lane_in_group];
+      if constexpr (GROUP >= 4) {
 #pragma unroll
-      for (int u = 0; u < GROUP; u += 4) consume4(mine, u);
+        for (int u = 0; u < GROUP; u += 4) consume4(mine, u);
+      } else {
+#pragma unroll
+        for (int u = 0; u < GROUP; ++u) consume(mine, u);
+      }
     }
     if (base < e) {  // remainder (< GROUP entries)
       const int32_t t = base + lane_in_group;

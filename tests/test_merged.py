@@ -67,11 +67,26 @@ def test_merged_structures_match_per_block(nb, w, first, last):
     X_stripe = (2 * rng.random(((last - first) * w, k)) - 1).astype(np.float32)
     X_0 = (2 * rng.random((w, k)) - 1).astype(np.float32)
 
-    # row-0 merged: C_0 = sum_c A_0c @ X_c
-    C0 = _apply_merged(eng._A_row0, X_stripe, X_0)
     C0_ref = np.zeros((w, k), np.float32)
     for j, c in enumerate(range(first, last)):
         C0_ref += eng.A_0i[j] @ X_stripe[j * w:(j + 1) * w]
+
+    if eng._A_all is not None:
+        # fully-fused single-process structure: one launch writes all rows
+        # (X_0 operand aliases the stripe head)
+        Call = _apply_merged(eng._A_all, X_stripe, X_stripe[:w])
+        np.testing.assert_allclose(Call[:w], C0_ref, rtol=1e-5, atol=1e-6)
+        for j, r in enumerate(range(first, last)):
+            if r == 0:
+                continue
+            ref = (eng.A_ii[j] @ X_stripe[j * w:(j + 1) * w]
+                   + eng.A_i0[j] @ X_stripe[:w])
+            np.testing.assert_allclose(Call[j * w:(j + 1) * w], ref,
+                                       rtol=1e-5, atol=1e-6)
+        return
+
+    # row-0 merged: C_0 = sum_c A_0c @ X_c
+    C0 = _apply_merged(eng._A_row0, X_stripe, X_0)
     np.testing.assert_allclose(C0, C0_ref, rtol=1e-5, atol=1e-6)
 
     # rest merged: C_r = A_rr @ X_r + A_r0 @ X_0
