@@ -172,10 +172,15 @@ class ArrowDecompositionMPI:
         for i in range(self.decomposition_length - 1):
             ex = self._forward[i]
             eng_s, eng_r = self.engines[i], self.engines[i + 1]
-            sendbuf = eng_s.backend.gather_rows(eng_s.feature_tile(), ex.send_rows)
-            recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
-            # C_i[recv_perm] = recvbuf; X := C (arrow_dec_mpi.py:544-545)
-            eng_r.backend.scatter_rows(eng_r.C_i, ex.recv_rows, recvbuf)
+            if self.comm.size == 1:
+                # rank-local: one fused permute pass, no staging buffers
+                eng_r.backend.permute_rows(eng_r.C_i, ex.recv_rows,
+                                           eng_s.feature_tile(), ex.send_rows)
+            else:
+                sendbuf = eng_s.backend.gather_rows(eng_s.feature_tile(), ex.send_rows)
+                recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
+                # C_i[recv_perm] = recvbuf; X := C (arrow_dec_mpi.py:544-545)
+                eng_r.backend.scatter_rows(eng_r.C_i, ex.recv_rows, recvbuf)
             eng_r.set_features(eng_r.C_i)
 
     def _aggregate(self) -> None:
@@ -184,10 +189,14 @@ class ArrowDecompositionMPI:
         for i in reversed(range(1, self.decomposition_length)):
             ex = self._backward[i]
             eng_s, eng_r = self.engines[i], self.engines[i - 1]
-            sendbuf = eng_s.backend.gather_rows(eng_s.result_tile(), ex.send_rows)
-            recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
-            # C_i[recv_perm] += recvbuf; X := C (arrow_dec_mpi.py:437-438)
-            eng_r.backend.scatter_add_rows(eng_r.C_i, ex.recv_rows, recvbuf)
+            if self.comm.size == 1:
+                eng_r.backend.permute_add_rows(eng_r.C_i, ex.recv_rows,
+                                               eng_s.result_tile(), ex.send_rows)
+            else:
+                sendbuf = eng_s.backend.gather_rows(eng_s.result_tile(), ex.send_rows)
+                recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
+                # C_i[recv_perm] += recvbuf; X := C (arrow_dec_mpi.py:437-438)
+                eng_r.backend.scatter_add_rows(eng_r.C_i, ex.recv_rows, recvbuf)
             eng_r.set_features(eng_r.C_i)
 
     # -- loading -------------------------------------------------------------

@@ -53,6 +53,12 @@ class CpuBackend:
     def scatter_add_rows(self, dst: torch.Tensor, idx: torch.Tensor, src: torch.Tensor):
         dst[idx] += src
 
+    def permute_rows(self, dst, dst_idx, src, src_idx):
+        dst[dst_idx] = src[src_idx]
+
+    def permute_add_rows(self, dst, dst_idx, src, src_idx):
+        dst[dst_idx] += src[src_idx]
+
     def synchronize(self):
         pass
 
@@ -118,6 +124,18 @@ class GpuBackend:
         if idx.shape[0]:
             hip.scatter_add_rows(dst.data_ptr(), src.data_ptr(), idx.data_ptr(),
                                  idx.shape[0], dst.shape[1], self._stream())
+
+    def permute_rows(self, dst, dst_idx, src, src_idx):
+        if dst_idx.shape[0]:
+            hip.permute_rows(dst.data_ptr(), src.data_ptr(), dst_idx.data_ptr(),
+                             src_idx.data_ptr(), dst_idx.shape[0], dst.shape[1],
+                             self._stream())
+
+    def permute_add_rows(self, dst, dst_idx, src, src_idx):
+        if dst_idx.shape[0]:
+            hip.permute_add_rows(dst.data_ptr(), src.data_ptr(),
+                                 dst_idx.data_ptr(), src_idx.data_ptr(),
+                                 dst_idx.shape[0], dst.shape[1], self._stream())
 
     def synchronize(self):
         torch.cuda.synchronize(self.device_index)

@@ -273,6 +273,57 @@ __global__ void route_rows_kernel(float *__restrict__ dst,
   }
 }
 
+// dst[dst_idx[i], :] (+)= src[src_idx[i], :] — fused gather+scatter for the
+// rank-local share of the permutation exchange (one read + one write per
+// element instead of a gather pass plus a scatter pass).
+template <typename VT, bool ADD>
+__global__ void permute_rows_kernel(float *__restrict__ dst,
+                                    const float *__restrict__ src,
+                                    const int64_t *__restrict__ dst_idx,
+                                    const int64_t *__restrict__ src_idx,
+                                    int64_t n, int64_t k_vec) {
+  const int64_t total = n * k_vec;
+  for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t i = t / k_vec, j = t % k_vec;
+    const VT v = reinterpret_cast<const VT *>(src)[src_idx[i] * k_vec + j];
+    VT *d = reinterpret_cast<VT *>(dst) + dst_idx[i] * k_vec + j;
+    if constexpr (!ADD) {
+      *d = v;
+    } else if constexpr (sizeof(VT) == 4) {
+      *d += v;
+    } else {
+      const float4 a = *reinterpret_cast<const float4 *>(&v);
+      float4 b = *reinterpret_cast<float4 *>(d);
+      b.x += a.x; b.y += a.y; b.z += a.z; b.w += a.w;
+      *reinterpret_cast<float4 *>(d) = b;
+    }
+  }
+}
+
+int launch_permute(int add, float *dst, const float *src, const int64_t *dst_idx,
+                   const int64_t *src_idx, int64_t n, int64_t k,
+                   hipStream_t stream) {
+  if (n == 0) return 0;
+  const bool vec4 = (k % 4 == 0);
+  const int64_t k_vec = vec4 ? k / 4 : k;
+  const int threads = 256;
+  int blocks = (int)std::min<int64_t>((n * k_vec + threads - 1) / threads, 16384);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, stream, dst, src,
+                       dst_idx, src_idx, n, k_vec);
+  };
+  if (vec4) {
+    if (add) launch(permute_rows_kernel<float4, true>);
+    else     launch(permute_rows_kernel<float4, false>);
+  } else {
+    if (add) launch(permute_rows_kernel<float, true>);
+    else     launch(permute_rows_kernel<float, false>);
+  }
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
 int launch_route(RouteOp op, float *dst, const float *src, const int64_t *idx,
                  int64_t n, int64_t k, hipStream_t stream) {
   if (n == 0) return 0;
@@ -516,6 +567,18 @@ int arrow_spmm_dual(int64_t handle, const float *X0_dev, const float *X1_dev,
 int arrow_spmm(int64_t handle, const float *X_dev, float *C_dev, int64_t k,
                int beta, void *stream_v) {
   return arrow_spmm_dual(handle, X_dev, X_dev, C_dev, k, beta, stream_v);
+}
+
+int arrow_permute_rows_f32(float *dst, const float *src, const int64_t *dst_idx,
+                           const int64_t *src_idx, int64_t n, int64_t k,
+                           void *stream) {
+  return launch_permute(0, dst, src, dst_idx, src_idx, n, k, (hipStream_t)stream);
+}
+
+int arrow_permute_add_rows_f32(float *dst, const float *src,
+                               const int64_t *dst_idx, const int64_t *src_idx,
+                               int64_t n, int64_t k, void *stream) {
+  return launch_permute(1, dst, src, dst_idx, src_idx, n, k, (hipStream_t)stream);
 }
 
 int arrow_gather_rows_f32(const float *src, float *dst, const int64_t *idx,

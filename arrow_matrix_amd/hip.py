@@ -57,6 +57,12 @@ def _load():
         fn.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                        ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p]
         fn.restype = ctypes.c_int
+    for name in ('arrow_permute_rows_f32', 'arrow_permute_add_rows_f32'):
+        fn = getattr(lib, name)
+        fn.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                       ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
+                       ctypes.c_void_p]
+        fn.restype = ctypes.c_int
     _lib = lib
     return lib
 
@@ -139,6 +145,18 @@ def scatter_rows(dst_ptr: int, src_ptr: int, idx_ptr: int, n: int, k: int, strea
 def scatter_add_rows(dst_ptr: int, src_ptr: int, idx_ptr: int, n: int, k: int, stream: int = 0):
     _check(_load().arrow_scatter_add_rows_f32(dst_ptr, src_ptr, idx_ptr, n, k, stream),
            "arrow_scatter_add_rows_f32")
+
+
+def permute_rows(dst_ptr, src_ptr, dst_idx_ptr, src_idx_ptr, n, k, stream=0):
+    _check(_load().arrow_permute_rows_f32(dst_ptr, src_ptr, dst_idx_ptr,
+                                          src_idx_ptr, n, k, stream),
+           "arrow_permute_rows_f32")
+
+
+def permute_add_rows(dst_ptr, src_ptr, dst_idx_ptr, src_idx_ptr, n, k, stream=0):
+    _check(_load().arrow_permute_add_rows_f32(dst_ptr, src_ptr, dst_idx_ptr,
+                                              src_idx_ptr, n, k, stream),
+           "arrow_permute_add_rows_f32")
 
 
 def abi_version() -> int:

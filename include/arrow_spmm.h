@@ -82,6 +82,16 @@ int arrow_scatter_add_rows_f32(float *dst_dev, const float *src_dev,
                                const int64_t *idx_dev, int64_t n, int64_t k,
                                void *stream);
 
+/* Fused gather+scatter for the rank-local share of the permutation routing
+ * (arrow_dec_mpi.py:526+544 / 421+437 collapsed into one pass):
+ * dst[dst_idx[i], :] (+)= src[src_idx[i], :] */
+int arrow_permute_rows_f32(float *dst_dev, const float *src_dev,
+                           const int64_t *dst_idx_dev, const int64_t *src_idx_dev,
+                           int64_t n, int64_t k, void *stream);
+int arrow_permute_add_rows_f32(float *dst_dev, const float *src_dev,
+                               const int64_t *dst_idx_dev, const int64_t *src_idx_dev,
+                               int64_t n, int64_t k, void *stream);
+
 #ifdef __cplusplus
 }
 #endif
