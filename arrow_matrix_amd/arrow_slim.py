@@ -205,9 +205,11 @@ class ArrowSlimMPI(ArrowMatrix):
                     self._A_bd_hi = (self.backend.upload_block(blk), local)
             n_rest += 1
         if fuse_all:
+            # x_rows = stripe only: the X_0 operand aliases the stripe head,
+            # so its rows are not extra algorithmic traffic
             self._A_all = self._merged_handle(nw * w, nw * w, rows_cat,
                                               cols_cat, data_cat,
-                                              x_rows=nw * w + w)
+                                              x_rows=nw * w)
         elif n_rest:
             rest_rows = nw * w - self._rest_row_offset
             self._A_rest = self._merged_handle(rest_rows, nw * w, rows_cat,
