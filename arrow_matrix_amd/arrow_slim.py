@@ -146,7 +146,13 @@ class ArrowSlimMPI(ArrowMatrix):
         self._rest_row_offset = 0
         if nw == 0:
             return
-        fuse_all = (self.comm.size == 1 and self.first_block == 0
+        # opt-in: measured a wash at 20M rows and -8% at 100M vs the
+        # two-launch layout (profiles/r01_*) — the X-stripe re-read the
+        # fusion saves is apparently absorbed by L3/scheduling overlap,
+        # while the merged structure loses locality
+        import os as _os
+        fuse_all = (_os.environ.get('ARROW_FUSE_ALL', '0') == '1'
+                    and self.comm.size == 1 and self.first_block == 0
                     and self.n_owned == self.tiles_per_side)
         # --- row-0 merge: C_0 = [A_0,first .. A_0,last-1] @ X_stripe -------
         rows_cat, cols_cat, data_cat = [], [], []
