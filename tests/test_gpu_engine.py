@@ -48,3 +48,25 @@ def test_native_library_is_loaded():
     hip._load()
     maps = open('/proc/self/maps').read()
     assert 'libarrowspmm.so' in maps
+
+
+@pytest.mark.gpu
+def test_bench_spmm_gpu_synthetic():
+    """The spmm_arrow CLI flow end-to-end on the GPU device (synthetic
+    fallback path, arrow_bench.py:28-41)."""
+    import os
+    import tempfile
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from arrow_matrix_amd.arrow_bench import bench_spmm
+    cwd = os.getcwd()
+    with tempfile.TemporaryDirectory() as td:
+        try:
+            os.chdir(td)
+            arrow = bench_spmm(None, 64, 8, 2, True, 'gpu',
+                               p_per_side=3, ba_neighbors=5)
+            assert arrow is not None
+            C = arrow.B.allgather_result()
+            assert np.isfinite(C).all() and np.abs(C).max() > 0
+        finally:
+            os.chdir(cwd)
