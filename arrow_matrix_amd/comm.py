@@ -39,6 +39,9 @@ class Comm:
     def allreduce_max_(self, tensor: torch.Tensor):
         pass
 
+    def allreduce_sum_(self, tensor: torch.Tensor, async_op: bool = False):
+        return None
+
     def alltoallv(self, send: torch.Tensor, send_counts: Sequence[int],
                   recv_counts: Sequence[int]) -> torch.Tensor:
         assert len(send_counts) == 1 and len(recv_counts) == 1
@@ -77,6 +80,10 @@ class TorchDistComm(Comm):
     def reduce_sum_(self, tensor: torch.Tensor, dst: int, async_op: bool = False):
         return dist.reduce(tensor, dst=dst, op=dist.ReduceOp.SUM,
                            group=self.group, async_op=async_op)
+
+    def allreduce_sum_(self, tensor: torch.Tensor, async_op: bool = False):
+        return dist.all_reduce(tensor, op=dist.ReduceOp.SUM, group=self.group,
+                               async_op=async_op)
 
     def allreduce_max_(self, tensor: torch.Tensor):
         # RCCL needs device tensors; round-trip CPU flags (e.g. the

@@ -227,6 +227,10 @@ def main():
 
     # features resident on device before the timed region
     eng0 = arrow.engines[0]
+    for eng in arrow.engines:
+        # iterated GNN-propagation loop (X := C): fuse C_0 reduce with the
+        # next X_0 broadcast into one allreduce
+        eng.allreduce_x0 = (args.device == 'gpu')
     if use_gpu:
         g = torch.Generator(device='cuda')
         g.manual_seed(42 + comm.rank)

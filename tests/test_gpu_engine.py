@@ -70,3 +70,43 @@ def test_bench_spmm_gpu_synthetic():
             assert np.isfinite(C).all() and np.abs(C).max() > 0
         finally:
             os.chdir(cwd)
+
+
+@pytest.mark.gpu
+def test_allreduce_x0_iterated_equivalence():
+    """The iterated-loop allreduce-X_0 optimisation (bench/cfg5 semantics)
+    gives the same results as the reduce+bcast path."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import os
+    import tempfile
+    from arrow_matrix_amd import graphio, synth
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+    from oracle import compute_spmm
+
+    n_blocks, width, k = [4], 64, 16
+    decomp = synth.synth_arrow_decomposition(width, n_blocks, avg_deg=6, seed=31)
+    n = n_blocks[0] * width
+    rng = np.random.default_rng(31)
+    X_orig = (2 * rng.random((n, k)) - 1).astype(np.float32)
+    with tempfile.TemporaryDirectory() as td:
+        prefix = os.path.join(td, 'g')
+        graphio.save_decomposition_new(decomp, prefix, width)
+        blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+            None, prefix, width)
+        arrow = ArrowDecompositionMPI.initialize(None, nb, tp, tn, width, k,
+                                                 device='gpu')
+        arrow.load_data_from_blocks(blocks)
+        arrow.zero_rhs(width, k)
+        for eng in arrow.engines:
+            eng.allreduce_x0 = True
+        perm0 = decomp[0][1]
+        arrow.B.set_features(X_orig[perm0].copy())
+        golden_X = X_orig
+        for _ in range(3):
+            arrow.step()
+            C = arrow.B.allgather_result()
+            golden = compute_spmm(decomp, golden_X)[perm0]
+            np.testing.assert_allclose(C, golden, rtol=5e-4, atol=5e-4)
+            golden_X = compute_spmm(decomp, golden_X)
+            arrow.B.set_features(arrow.B.result_tile())
