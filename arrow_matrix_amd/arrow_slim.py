@@ -144,6 +144,25 @@ class ArrowSlimMPI(ArrowMatrix):
             # drop the host block references — device copies are resident
             self.A_0i = self.A_ii = self.A_i0 = None
             self.A_lo = self.A_hi = None
+        else:
+            self._optimize_Ai_slices()
+
+    def _optimize_Ai_slices(self, threshold: float = 0.3) -> None:
+        """Column-compact cpu blocks with < threshold nonzero columns,
+        keeping the nnz-column map used to gather X at multiply time
+        (reference arrow_slim_mpi.py:329-350). On the GPU this is moot: the
+        resident merged structure touches only the X rows its indices name,
+        so there is nothing to compact."""
+        self._nnz_columns = [[None, None, None] for _ in range(max(self.n_owned, 1))]
+        for j in range(self.n_owned):
+            for slot, blocks in ((0, self.A_0i), (1, self.A_ii), (2, self.A_i0)):
+                b = blocks[j]
+                if b is None:
+                    continue
+                nnz_cols = np.unique(b.nonzero()[1])
+                if len(nnz_cols) < threshold * b.shape[1]:
+                    blocks[j] = b[:, nnz_cols]
+                    self._nnz_columns[j][slot] = nnz_cols
 
     def _build_merged_gpu(self) -> None:
         w = self.width
@@ -457,10 +476,14 @@ class ArrowSlimMPI(ArrowMatrix):
         be = self.backend
         w = self.width
 
+        def xsel(X, j, slot):
+            cols = self._nnz_columns[j][slot] if hasattr(self, '_nnz_columns') else None
+            return X.contiguous() if cols is None else X[cols].contiguous()
+
         first = True
         for j, r in enumerate(range(self.first_block, self.last_block)):
             Xr = self.X_i[j * w:(j + 1) * w]
-            be.spmm_block(self.A_0i[j], Xr.contiguous(), self.C_0, 0 if first else 1)
+            be.spmm_block(self.A_0i[j], xsel(Xr, j, 0), self.C_0, 0 if first else 1)
             first = False
         if first:  # rank owns no blocks of this matrix
             self.C_0.zero_()
@@ -472,8 +495,8 @@ class ArrowSlimMPI(ArrowMatrix):
                 continue
             Xr = self.X_i[j * w:(j + 1) * w]
             Cr = self.C_i[j * w:(j + 1) * w]
-            be.spmm_block(self.A_ii[j], Xr.contiguous(), Cr, 0)
-            be.spmm_block(self.A_i0[j], self.X_0, Cr, 1)
+            be.spmm_block(self.A_ii[j], xsel(Xr, j, 1), Cr, 0)
+            be.spmm_block(self.A_i0[j], xsel(self.X_0, j, 2), Cr, 1)
             # banded ±1 halo terms (arrow_mpi.py:211-219)
             lo = self.A_lo[j] if self.A_lo else None
             hi = self.A_hi[j] if self.A_hi else None
