@@ -67,6 +67,9 @@ struct CsrBlock {
   int64_t n_items = 0;
   int32_t *split_rows = nullptr;  // device: rows needing pre-zero at beta=0
   int64_t n_split_rows = 0;
+  // XCD-contiguous item remap for this structure (uniform banded rows only:
+  // on hub-heavy structures it serialises the heavy head onto one XCD)
+  int xcd_remap = 0;
 };
 
 std::unordered_map<int64_t, CsrBlock> g_blocks;
@@ -382,16 +385,10 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
   for (int64_t col_off = 0; col_off < k; col_off += span) {
     const bool guard = (col_off + span > k);
     auto run = [&](auto kern) {
-      // default OFF: the contiguous-per-XCD remap measured 16% SLOWER on
-      // the cfg4-shaped workload (profiles/r01_*) — the shared moving
-      // window over L3 beats 8 disjoint per-XCD windows here
-      static const int xcd_remap =
-          [] { const char *e = getenv("ARROW_SPMM_XCD_REMAP");
-               return (e && e[0] == '1') ? 1 : 0; }();
       hipLaunchKernelGGL(kern, dim3(blocks), dim3(BLOCK_THREADS), 0, stream,
                          blk.pairs, blk.item_row, blk.item_begin,
                          blk.item_end, blk.n_items, X0, X1, C, k, col_off,
-                         xcd_remap);
+                         blk.xcd_remap);
     };
     if (beta == 0) {
       if (guard) run(spmm_kernel<VEC, GROUP, 0, true>);
@@ -531,6 +528,16 @@ int64_t arrow_csr_nnz(int64_t handle) {
   auto it = g_blocks.find(handle);
   if (it == g_blocks.end()) return -1;
   return it->second.nnz;
+}
+
+int arrow_csr_set_xcd_remap(int64_t handle, int enable) {
+  auto it = g_blocks.find(handle);
+  if (it == g_blocks.end()) {
+    set_error("arrow_csr_set_xcd_remap: bad handle");
+    return -1;
+  }
+  it->second.xcd_remap = enable ? 1 : 0;
+  return 0;
 }
 
 int arrow_spmm_dual(int64_t handle, const float *X0_dev, const float *X1_dev,

@@ -44,6 +44,8 @@ def _load():
     lib.arrow_csr_destroy.restype = ctypes.c_int
     lib.arrow_csr_nnz.argtypes = [ctypes.c_int64]
     lib.arrow_csr_nnz.restype = ctypes.c_int64
+    lib.arrow_csr_set_xcd_remap.argtypes = [ctypes.c_int64, ctypes.c_int]
+    lib.arrow_csr_set_xcd_remap.restype = ctypes.c_int
     lib.arrow_spmm.argtypes = [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
                                ctypes.c_int64, ctypes.c_int, ctypes.c_void_p]
     lib.arrow_spmm.restype = ctypes.c_int
@@ -108,6 +110,10 @@ class CsrBlockGPU:
         """C (+)= A @ X on device pointers (e.g. torch tensor data_ptr())."""
         _check(_load().arrow_spmm(self._handle, X_ptr, C_ptr, k, beta, stream),
                "arrow_spmm")
+
+    def set_xcd_remap(self, enable: bool):
+        _check(_load().arrow_csr_set_xcd_remap(self._handle, 1 if enable else 0),
+               "arrow_csr_set_xcd_remap")
 
     def spmm_dual(self, X0_ptr: int, X1_ptr: int, C_ptr: int, k: int,
                   beta: int, stream: int = 0):

@@ -55,6 +55,9 @@ int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
 int arrow_csr_destroy(int64_t handle);
 /* nnz of a resident block (for flop accounting) */
 int64_t arrow_csr_nnz(int64_t handle);
+/* Enable the XCD-contiguous work remap for this structure (a performance
+ * hint for uniform banded rows; off by default). */
+int arrow_csr_set_xcd_remap(int64_t handle, int enable);
 
 /* C (+)= A @ X.  X: (cols, k) fp32 row-major device;  C: (rows, k).
  * beta = 0: C = A@X (rows not touched by A are zeroed);  beta = 1: C += A@X. */
