@@ -285,6 +285,7 @@ class ArrowSlimMPI(ArrowMatrix):
                     setattr(self, name, self.backend.zeros((w, k)))
                 else:
                     buf.zero_()
+        self._x0_valid = False  # buffers were just reset
         # ping-pong pool: spmm writes C into a stripe that does NOT alias
         # X_i, so `X := C` between iterations (set_features(result_tile()))
         # is race-free on the GPU (the reference allocates a fresh C every

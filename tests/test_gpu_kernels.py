@@ -43,7 +43,7 @@ def _check_spmm(gpu, A, k, beta, seed=0, rtol=1e-5, atol=1e-5):
     np.testing.assert_allclose(got, ref, rtol=rtol, atol=atol * scale)
 
 
-@pytest.mark.parametrize("k", [1, 2, 3, 4, 5, 16, 32, 64, 128])
+@pytest.mark.parametrize("k", [1, 2, 3, 4, 5, 6, 8, 12, 16, 24, 32, 64, 128])
 def test_spmm_k_sweep(gpu, k):
     A = _random_csr(500, 700, 0.02, seed=k)
     _check_spmm(gpu, A, k, beta=0, seed=k)
