@@ -48,6 +48,10 @@ class Comm:
         assert send_counts[0] == recv_counts[0]
         return send
 
+    def alltoallv_async(self, send: torch.Tensor, send_counts, recv_counts):
+        assert len(send_counts) == 1 and len(recv_counts) == 1
+        return send, []
+
     def allgather_cat(self, tensor: torch.Tensor) -> torch.Tensor:
         return tensor
 
@@ -106,9 +110,6 @@ class TorchDistComm(Comm):
         The self-block is copied locally (never hits the backend).
         """
         assert send.dim() == 2
-        k = send.shape[1]
-        total_recv = int(sum(recv_counts))
-        recv = send.new_empty((total_recv, k))
         sdispl = [0]
         for c in send_counts:
             sdispl.append(sdispl[-1] + int(c))
@@ -116,6 +117,27 @@ class TorchDistComm(Comm):
         for c in recv_counts:
             rdispl.append(rdispl[-1] + int(c))
 
+        recv, works = self._post_alltoallv(send, send_counts, recv_counts,
+                                           sdispl, rdispl)
+        for w in works:
+            w.wait()
+        return recv
+
+    def alltoallv_async(self, send: torch.Tensor, send_counts, recv_counts):
+        """Post the exchange and return (recv, works) WITHOUT waiting —
+        kernels enqueued before the waits overlap with the transfer
+        (RCCL runs on its own stream; wait() inserts stream deps)."""
+        sdispl = [0]
+        for c in send_counts:
+            sdispl.append(sdispl[-1] + int(c))
+        rdispl = [0]
+        for c in recv_counts:
+            rdispl.append(rdispl[-1] + int(c))
+        return self._post_alltoallv(send, send_counts, recv_counts, sdispl, rdispl)
+
+    def _post_alltoallv(self, send, send_counts, recv_counts, sdispl, rdispl):
+        k = send.shape[1]
+        recv = send.new_empty((int(sum(recv_counts)), k))
         ops = []
         for r in range(self.size):
             if r == self.rank:
@@ -127,14 +149,12 @@ class TorchDistComm(Comm):
             if recv_counts[r] > 0:
                 ops.append(dist.P2POp(dist.irecv, recv[rdispl[r]:rdispl[r + 1]],
                                       r, group=self.group))
-        if ops:
-            for w in dist.batch_isend_irecv(ops):
-                w.wait()
-        # local block
+        works = dist.batch_isend_irecv(ops) if ops else []
+        # local block (current-stream device copy; overlaps with NCCL)
         if send_counts[self.rank] > 0:
             recv[rdispl[self.rank]:rdispl[self.rank + 1]].copy_(
                 send[sdispl[self.rank]:sdispl[self.rank] + send_counts[self.rank]])
-        return recv
+        return recv, works
 
     def allgather_cat(self, tensor: torch.Tensor) -> torch.Tensor:
         out: List[torch.Tensor] = [torch.empty_like(tensor) for _ in range(self.size)]
