@@ -109,12 +109,28 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
   int64_t item = (int64_t)wg * GROUPS_PER_BLOCK + group_in_block;
   const int64_t stride = (int64_t)gridDim.x * GROUPS_PER_BLOCK;
 
+  // software prefetch of the NEXT item's metadata: without it every ~8-nnz
+  // item pays a 3-deep dependent load chain (meta -> pairs -> X row) that
+  // dominates short power-law rows
+  int32_t next_row = 0, next_b = 0, next_e = 0;
+  if (item < n_items) {
+    next_row = item_row[item];
+    next_b = item_begin[item];
+    next_e = item_end[item];
+  }
+
   for (; item < n_items; item += stride) {
-    const int32_t row_raw = item_row[item];
+    const int32_t row_raw = next_row;
     const int32_t row = row_raw & 0x7fffffff;
     const bool is_split = row_raw < 0;
-    const int32_t b = item_begin[item];
-    const int32_t e = item_end[item];
+    const int32_t b = next_b;
+    const int32_t e = next_e;
+    const int64_t nxt = item + stride;
+    if (nxt < n_items) {
+      next_row = item_row[nxt];
+      next_b = item_begin[nxt];
+      next_e = item_end[nxt];
+    }
 
     float acc[VEC];
 #pragma unroll
@@ -367,6 +383,10 @@ struct LaunchCfg {
 };
 
 LaunchCfg pick_cfg(int64_t k) {
+  static const int g64 =
+      [] { const char *e = getenv("ARROW_SPMM_G64");
+           return (e && e[0] == '1') ? 1 : 0; }();
+  if (g64 && k == 128) return {2, 64, 128};  // A/B: wave-wide groups, float2
   int vec = (k % 4 == 0) ? 4 : (k % 2 == 0) ? 2 : 1;
   int64_t lanes_needed = (k + vec - 1) / vec;
   int group = 1;
