@@ -25,7 +25,7 @@ MI355X-first deviations:
 from typing import List, Optional
 
 import numpy as np
-import torch
+
 
 from . import graphio, tables
 from .arrow_matrix import ArrowMatrix

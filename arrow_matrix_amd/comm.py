@@ -11,7 +11,7 @@ One process per GPU; all payloads are torch tensors (CUDA tensors under
 nccl, CPU tensors under gloo — so the same engine code runs the gloo
 world_size>1 CPU tests and the RCCL GPU path).
 """
-from typing import List, Optional, Sequence
+from typing import List, Sequence
 
 import torch
 
