@@ -10,8 +10,11 @@ setup(
     python_requires='>=3.10',
     entry_points={
         'console_scripts': [
-            # same entry point name as the reference (setup.py:20)
+            # same entry point names as the reference (setup.py:17-24)
+            'arrow_decompose = scripts.decomposition_main:main',
             'spmm_arrow = scripts.spmm_arrow_main:main',
+            'spmm_15d = scripts.spmm_15d_main:main',
+            'spmm_petsc = scripts.spmm_petsc_main:main',
         ],
     },
 )
