@@ -197,10 +197,43 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
         consume(mine, u + 2); consume(mine, u + 3);
       }
     };
+    // 8 loads in flight: a deg-8 power-law row issues its entire X-row
+    // load set before any use (the 4-deep batch still serialises pairs of
+    // batches on in-order issue)
+    auto consume8 = [&](int2 mine, int u) {
+      if constexpr (VEC == 4 && GROUP >= 8) {
+        const float *p[8];
+        float v[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          p[j] = xaddr(mine, u + j);
+          v[j] = val(mine, u + j);
+        }
+        if (active) {
+          float4 a[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            a[j] = *reinterpret_cast<const float4 *>(p[j]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            acc[0] = fmaf(v[j], a[j].x, acc[0]);
+            acc[1] = fmaf(v[j], a[j].y, acc[1]);
+            acc[2] = fmaf(v[j], a[j].z, acc[2]);
+            acc[3] = fmaf(v[j], a[j].w, acc[3]);
+          }
+        }
+      } else {
+        consume4(mine, u);
+        consume4(mine, u + 4);
+      }
+    };
     int32_t base = b;
     for (; base + GROUP <= e; base += GROUP) {  // full chunks
       const int2 mine = pairs[base + lane_in_group];
-      if constexpr (GROUP >= 4) {
+      if constexpr (GROUP >= 8) {
+#pragma unroll
+        for (int u = 0; u < GROUP; u += 8) consume8(mine, u);
+      } else if constexpr (GROUP >= 4) {
 #pragma unroll
         for (int u = 0; u < GROUP; u += 4) consume4(mine, u);
       } else {
@@ -213,6 +246,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
       const int2 mine = (t < e) ? pairs[t] : int2{0, 0};
       const int cnt = e - base;
       int u = 0;
+      if constexpr (GROUP >= 8) {
+        for (; u + 8 <= cnt; u += 8) consume8(mine, u);
+      }
       for (; u + 4 <= cnt; u += 4) consume4(mine, u);
       for (; u < cnt; ++u) consume(mine, u);
     }
