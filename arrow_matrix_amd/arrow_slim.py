@@ -169,6 +169,7 @@ class ArrowSlimMPI(ArrowMatrix):
         nw = self.n_owned
         self._A_row0 = None
         self._A_rest = None
+        self._A_all = None
         self._rest_row_offset = 0
         if nw == 0:
             return
