@@ -75,6 +75,7 @@ class ArrowSlimMPI(ArrowMatrix):
         self._A_bd_lo = None          # boundary (first, first-1) vs halo_lo
         self._A_bd_hi = None          # boundary (last-1, last) vs halo_hi
         self._A_all = None            # single-process fully-fused structure
+        self._A_col = None            # A/B: X_0-column entries, hub-sorted
         # iterated-loop optimisation (bench/cfg5 semantics, X := C between
         # steps): replace iteration t's C_0 Reduce + iteration t+1's X_0
         # Bcast with ONE allreduce — next X_0 IS the reduced C_0. Off by
@@ -210,6 +211,14 @@ class ArrowSlimMPI(ArrowMatrix):
             rows_cat, cols_cat, data_cat = row0_sets
         else:
             rows_cat, cols_cat, data_cat = [], [], []
+        # A/B (ARROW_SPLIT_COL=1): keep the A_r0 (X_0) entries OUT of the
+        # rest structure and process them in a separate launch whose row
+        # order is sorted by hub column — the zipf-skewed X_0 gather then
+        # walks the hot head near-sequentially (cache-friendly) at the cost
+        # of one extra C read-modify-write pass
+        split_col = (_os.environ.get('ARROW_SPLIT_COL', '0') == '1'
+                     and not fuse_all)
+        col_rows, col_cols, col_data = [], [], []
         n_rest = 0
         for j, r in enumerate(range(self.first_block, self.last_block)):
             if r == 0:
@@ -220,9 +229,14 @@ class ArrowSlimMPI(ArrowMatrix):
             rows_cat.append(local + np.repeat(np.arange(w), np.diff(rr.indptr)))
             cols_cat.append(rr.indices.astype(np.int64) + j * w)
             data_cat.append(rr.data.astype(np.float32))
-            rows_cat.append(local + np.repeat(np.arange(w), np.diff(r0.indptr)))
-            cols_cat.append(-(r0.indices.astype(np.int64) + 1))  # -> X_0
-            data_cat.append(r0.data.astype(np.float32))
+            if split_col:
+                col_rows.append(local + np.repeat(np.arange(w), np.diff(r0.indptr)))
+                col_cols.append(-(r0.indices.astype(np.int64) + 1))  # -> X_0
+                col_data.append(r0.data.astype(np.float32))
+            else:
+                rows_cat.append(local + np.repeat(np.arange(w), np.diff(r0.indptr)))
+                cols_cat.append(-(r0.indices.astype(np.int64) + 1))  # -> X_0
+                data_cat.append(r0.data.astype(np.float32))
             for off, blk in ((-1, lo_list[j]), (1, hi_list[j])):
                 if blk is None:
                     continue
@@ -247,7 +261,40 @@ class ArrowSlimMPI(ArrowMatrix):
             rest_rows = nw * w - self._rest_row_offset
             self._A_rest = self._merged_handle(rest_rows, nw * w, rows_cat,
                                                cols_cat, data_cat,
-                                               x_rows=n_rest * w + w)
+                                               x_rows=(n_rest * w + w
+                                                       if not split_col else n_rest * w))
+            if split_col and col_rows:
+                self._A_col = self._merged_col_sorted(rest_rows, nw * w,
+                                                      col_rows, col_cols,
+                                                      col_data)
+
+    def _merged_col_sorted(self, n_rows, n_cols, rows_cat, cols_cat, data_cat):
+        """Build the X_0-entry structure with ROW BLOCKS ordered by their
+        hottest (smallest) hub column; each row appears once (exclusive C
+        writes via explicit row ids), so the launch is a plain beta=1 pass."""
+        rows = np.concatenate(rows_cat)
+        cols = np.concatenate(cols_cat)
+        data = np.concatenate(data_cat)
+        x0col = -cols - 1
+        order = np.lexsort((x0col, rows))  # group by row, min col first
+        rows, cols, data, x0col = rows[order], cols[order], data[order], x0col[order]
+        uniq_rows, starts = np.unique(rows, return_index=True)
+        key = x0col[starts]                      # per-row min hub column
+        row_rank = np.argsort(key, kind='stable')
+        perm_rows = uniq_rows[row_rank]          # output rows in key order
+        remap = np.full(n_rows, -1, dtype=np.int64)
+        remap[perm_rows] = np.arange(perm_rows.size)
+        new_rows = remap[rows]
+        order2 = np.argsort(new_rows, kind='stable')
+        cols2, data2 = cols[order2], data[order2]
+        counts = np.bincount(new_rows, minlength=perm_rows.size)
+        indptr = np.zeros(perm_rows.size + 1, dtype=np.int64)
+        np.cumsum(counts, out=indptr[1:])
+        handle = self.backend.upload_arrays(
+            (perm_rows.size, n_cols), indptr, cols2.astype(np.int32),
+            data2.astype(np.float32), row_ids=perm_rows.astype(np.int64))
+        handle.x_rows = self.width
+        return handle
 
     def _merged_handle(self, n_rows, n_cols, rows_cat, cols_cat, data_cat,
                        x_rows):
@@ -431,6 +478,12 @@ class ArrowSlimMPI(ArrowMatrix):
                              self.n_owned * w]
             self._timed(lambda: be.spmm_dual(h, self.X_i, self.X_0, C_sub, 0),
                         h.nnz, C_sub.shape[0], h.x_rows)
+        if self._A_col is not None:
+            h = self._A_col
+            C_full = self.C_i[self._rest_row_offset:self.n_owned * w]
+            self._timed(lambda: be.spmm_dual(h, self.X_i, self.X_0, C_full, 1),
+                        h.nnz, h.shape[0], h.x_rows)
+
         # banded boundary off-diagonals against the received halo tiles
         for bd, halo in ((self._A_bd_lo, self.X_halo_lo),
                          (self._A_bd_hi, self.X_halo_hi)):

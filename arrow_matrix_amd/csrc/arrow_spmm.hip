@@ -500,9 +500,9 @@ int arrow_synchronize(void) {
   return 0;
 }
 
-int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
-                         const int64_t *indptr, const int32_t *indices,
-                         const float *data) {
+static int64_t csr_create_impl(int64_t rows, int64_t cols, int64_t nnz,
+                               const int64_t *indptr, const int32_t *indices,
+                               const float *data, const int64_t *row_ids) {
   if (rows < 0 || cols < 0 || nnz < 0 || (rows > 0 && !indptr)) {
     set_error("arrow_csr_create: bad arguments");
     return -1;
@@ -522,14 +522,17 @@ int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
   item_row.reserve(rows + nnz / SEG_NNZ + 1);
   for (int64_t r = 0; r < rows; ++r) {
     const int64_t b = indptr[r], e = indptr[r + 1];
+    // row_ids (optional) maps structure-row -> output C row, enabling
+    // reordered layouts (e.g. the hub-sorted X_0 structure)
+    const int32_t out_row = (int32_t)(row_ids ? row_ids[r] : r);
     if (e - b <= SEG_NNZ) {
-      item_row.push_back((int32_t)r);
+      item_row.push_back(out_row);
       item_begin.push_back((int32_t)b);
       item_end.push_back((int32_t)e);
     } else {
-      split_rows.push_back((int32_t)r);
+      split_rows.push_back(out_row);
       for (int64_t s = b; s < e; s += SEG_NNZ) {
-        item_row.push_back((int32_t)r | INT32_MIN);
+        item_row.push_back(out_row | INT32_MIN);
         item_begin.push_back((int32_t)s);
         item_end.push_back((int32_t)std::min<int64_t>(s + SEG_NNZ, e));
       }
@@ -562,6 +565,18 @@ int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
   const int64_t h = g_next_handle++;
   g_blocks.emplace(h, blk);
   return h;
+}
+
+int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
+                         const int64_t *indptr, const int32_t *indices,
+                         const float *data) {
+  return csr_create_impl(rows, cols, nnz, indptr, indices, data, nullptr);
+}
+
+int64_t arrow_csr_create_rows(int64_t rows, int64_t cols, int64_t nnz,
+                              const int64_t *indptr, const int32_t *indices,
+                              const float *data, const int64_t *row_ids) {
+  return csr_create_impl(rows, cols, nnz, indptr, indices, data, row_ids);
 }
 
 int arrow_csr_destroy(int64_t handle) {

@@ -40,6 +40,11 @@ def _load():
         ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
         ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
         ctypes.POINTER(ctypes.c_float)]
+    lib.arrow_csr_create_rows.restype = ctypes.c_int64
+    lib.arrow_csr_create_rows.argtypes = [
+        ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
+        ctypes.POINTER(ctypes.c_float), ctypes.POINTER(ctypes.c_int64)]
     lib.arrow_csr_destroy.argtypes = [ctypes.c_int64]
     lib.arrow_csr_destroy.restype = ctypes.c_int
     lib.arrow_csr_nnz.argtypes = [ctypes.c_int64]
@@ -83,10 +88,11 @@ class CsrBlockGPU:
     """A CSR block resident in HBM (uploaded ONCE — unlike the reference,
     which re-uploads A every iteration, arrow_slim_mpi.py:184-232)."""
 
-    def __init__(self, csr=None, arrays=None):
+    def __init__(self, csr=None, arrays=None, row_ids=None):
         """csr: scipy CSR, or arrays=(shape, indptr, indices, data) for raw
         uploads (the fused layouts use negative column indices, which scipy
-        would reject)."""
+        would reject). row_ids: optional explicit output-row id per
+        structure row (reordered layouts)."""
         lib = _load()
         if arrays is not None:
             (rows, cols), indptr, indices, data = arrays
@@ -99,12 +105,22 @@ class CsrBlockGPU:
         data = np.ascontiguousarray(data, dtype=np.float32)
         self.shape = (rows, cols)
         self.nnz = int(indices.size)
-        self._handle = _check(lib.arrow_csr_create(
-            rows, cols, self.nnz,
-            indptr.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
-            indices.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
-            data.ctypes.data_as(ctypes.POINTER(ctypes.c_float))),
-            "arrow_csr_create")
+        if row_ids is not None:
+            row_ids = np.ascontiguousarray(row_ids, dtype=np.int64)
+            self._handle = _check(lib.arrow_csr_create_rows(
+                rows, cols, self.nnz,
+                indptr.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+                indices.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+                data.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+                row_ids.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))),
+                "arrow_csr_create_rows")
+        else:
+            self._handle = _check(lib.arrow_csr_create(
+                rows, cols, self.nnz,
+                indptr.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+                indices.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+                data.ctypes.data_as(ctypes.POINTER(ctypes.c_float))),
+                "arrow_csr_create")
 
     def spmm(self, X_ptr: int, C_ptr: int, k: int, beta: int, stream: int = 0):
         """C (+)= A @ X on device pointers (e.g. torch tensor data_ptr())."""

@@ -52,6 +52,11 @@ int arrow_abi_version(void);
 int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
                          const int64_t *indptr, const int32_t *indices,
                          const float *data);
+/* As arrow_csr_create, but structure row r writes output row row_ids[r]
+ * (reordered layouts, e.g. hub-sorted first-block-column entries). */
+int64_t arrow_csr_create_rows(int64_t rows, int64_t cols, int64_t nnz,
+                              const int64_t *indptr, const int32_t *indices,
+                              const float *data, const int64_t *row_ids);
 int arrow_csr_destroy(int64_t handle);
 /* nnz of a resident block (for flop accounting) */
 int64_t arrow_csr_nnz(int64_t handle);

@@ -110,3 +110,19 @@ def test_allreduce_x0_iterated_equivalence():
             np.testing.assert_allclose(C, golden, rtol=5e-4, atol=5e-4)
             golden_X = compute_spmm(decomp, golden_X)
             arrow.B.set_features(arrow.B.result_tile())
+
+
+@pytest.mark.gpu
+def test_engine_gpu_split_col(monkeypatch):
+    """ARROW_SPLIT_COL=1 (hub-sorted X_0 structure) parity on GPU."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    monkeypatch.setenv('ARROW_SPLIT_COL', '1')
+    from tests.test_engine_cpu import _run_engine
+    from arrow_matrix_amd import synth
+    decomp = synth.synth_arrow_decomposition(64, [4], avg_deg=8, seed=17,
+                                             hub_rows=2)
+    results, goldens = _run_engine(decomp, 64, [4], 16, iters=2,
+                                   device='gpu', seed=17)
+    for C, G in zip(results, goldens):
+        np.testing.assert_allclose(C, G, rtol=2e-4, atol=2e-4)

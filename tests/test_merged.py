@@ -18,10 +18,11 @@ class _CaptureBackend:
     def __init__(self):
         self.uploads = []
 
-    def upload_arrays(self, shape, indptr, indices, data):
+    def upload_arrays(self, shape, indptr, indices, data, row_ids=None):
         h = SimpleNamespace(shape=shape, indptr=np.asarray(indptr),
                             indices=np.asarray(indices),
-                            data=np.asarray(data), nnz=int(len(indices)))
+                            data=np.asarray(data), nnz=int(len(indices)),
+                            row_ids=None if row_ids is None else np.asarray(row_ids))
         self.uploads.append(h)
         return h
 
@@ -100,3 +101,43 @@ def test_merged_structures_match_per_block(nb, w, first, last):
                    + eng.A_i0[j] @ X_0)
             got = Crest[j * w - off:(j + 1) * w - off]
             np.testing.assert_allclose(got, ref, rtol=1e-5, atol=1e-6)
+
+
+def test_split_col_structure(monkeypatch):
+    """ARROW_SPLIT_COL=1: the hub-sorted X_0 structure + rest structure
+    together equal the unsplit computation."""
+    monkeypatch.setenv('ARROW_SPLIT_COL', '1')
+    nb, w, first, last = 4, 5, 1, 4   # no block 0 owned -> offset 0
+    decomp = synth.synth_arrow_decomposition(w, [nb], avg_deg=5, seed=3)
+    B, _ = decomp[0]
+    blocks = split_matrix_to_blocks(B, w)
+    eng = ArrowSlimMPI(None, tiles_per_side=nb, device='cpu')
+    eng.width = w
+    eng.first_block, eng.last_block = first, last
+    eng.n_owned = last - first
+    eng.A_0i = [sparse.csr_matrix(blocks[0][c]) for c in range(first, last)]
+    eng.A_ii = [sparse.csr_matrix(blocks[r][r]) for r in range(first, last)]
+    eng.A_i0 = [sparse.csr_matrix(blocks[r][0]) for r in range(first, last)]
+    cap = _CaptureBackend()
+    eng.backend = cap
+    eng._build_merged_gpu()
+    assert eng._A_col is not None
+    # hub-sort property: per-structure-row min X_0 col is non-decreasing
+    h = eng._A_col
+    mins = [min(-h.indices[h.indptr[r]:h.indptr[r+1]] - 1)
+            for r in range(h.shape[0]) if h.indptr[r+1] > h.indptr[r]]
+    assert all(mins[i] <= mins[i+1] for i in range(len(mins)-1))
+
+    rng = np.random.default_rng(0)
+    k = 3
+    X_stripe = (2 * rng.random(((last - first) * w, k)) - 1).astype(np.float32)
+    X_0 = (2 * rng.random((w, k)) - 1).astype(np.float32)
+    C = _apply_merged(eng._A_rest, X_stripe, X_0)
+    # add the col structure with its explicit row ids
+    for r in range(h.shape[0]):
+        out = int(h.row_ids[r])
+        for t in range(h.indptr[r], h.indptr[r+1]):
+            C[out] += h.data[t] * X_0[-int(h.indices[t]) - 1]
+    for j, r in enumerate(range(first, last)):
+        ref = eng.A_ii[j] @ X_stripe[j*w:(j+1)*w] + eng.A_i0[j] @ X_0
+        np.testing.assert_allclose(C[j*w:(j+1)*w], ref, rtol=1e-5, atol=1e-6)
