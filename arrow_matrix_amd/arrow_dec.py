@@ -305,8 +305,14 @@ def _extract_block(A, i: int, j: int, block_size: int):
     c0, c1 = j * block_size, min(cols, (j + 1) * block_size)
     sl = A[r0:r1, c0:c1]
     pad_width = block_size - (r1 - r0)
+    # Always square (block_size, block_size): the reference pads only short
+    # last block-ROWS (graphio.py:394-399) and asserts squareness at load
+    # (arrow_slim_mpi.py:305), which breaks on a ragged last block-COLUMN;
+    # declaring the full square shape (extra rows/cols are empty) keeps the
+    # engine's uniform tiles with identical results.
     if pad_width == 0:
-        block = sparse.csr_matrix(sl)
+        block = sparse.csr_matrix((sl.data, sl.indices, sl.indptr),
+                                  shape=(block_size, block_size))
     else:
         indx_ptr = np.pad(sl.indptr, (0, pad_width), mode='edge')
         block = sparse.csr_matrix((sl.data, sl.indices, indx_ptr),

@@ -128,3 +128,41 @@ def test_end_to_end_decompose_then_engine():
         golden = (A @ X_orig)[perm0]
         np.testing.assert_allclose(C[:min(n_pad, n)], golden[:min(n_pad, n)],
                                    rtol=1e-4, atol=1e-4)
+
+
+def test_end_to_end_ragged_tail():
+    """n not a multiple of width: the loader's edge-padded square last
+    block (graphio.py:394-399) through producer -> files -> engine."""
+    import os
+    import tempfile
+    from arrow_matrix_amd import graphio
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+
+    n, width = 130, 24   # ceil(130/24) = 6 blocks, last block 10 rows ragged
+    A = _random_graph(n, 6, 21, power_law=True)
+    decomp = arrow_decomposition(A, width, max_number_of_levels=4,
+                                 block_diagonal=True,
+                                 rng=np.random.default_rng(4))
+    pairs = [(p.graph, p.permutation) for p in decomp]
+    with tempfile.TemporaryDirectory() as td:
+        prefix = os.path.join(td, 'g')
+        graphio.save_decomposition_new(pairs, prefix, width)
+        blocks, nb, to_prev, to_next = ArrowDecompositionMPI.load_decomposition_new(
+            None, prefix, width)
+        k = 5
+        arrow = ArrowDecompositionMPI.initialize(None, nb, to_prev, to_next,
+                                                 width, k, device='cpu')
+        arrow.load_data_from_blocks(blocks)
+        arrow.zero_rhs(width, k)
+        rng = np.random.default_rng(5)
+        n_pad = int(nb[0]) * width
+        X_orig = (2 * rng.random((n, k)) - 1).astype(np.float32)
+        perm0 = np.asarray(decomp[0].permutation)
+        X_eng = np.zeros((n_pad, k), np.float32)
+        m = min(n_pad, n)
+        X_eng[:m] = X_orig[perm0][:m]
+        arrow.B.set_features(X_eng.copy())
+        arrow.step()
+        C = arrow.B.allgather_result()
+        golden = (A @ X_orig)[perm0]
+        np.testing.assert_allclose(C[:m], golden[:m], rtol=1e-4, atol=1e-4)
