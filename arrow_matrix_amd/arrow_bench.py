@@ -68,6 +68,7 @@ def bench_spmm(path: Union[str, None],
     arrow = ArrowDecompositionMPI.initialize(comm, n_blocks, to_prev, to_next,
                                              width, n_features, device, blocked, slim)
 
+    wb_logging.log({"actual_ranks": comm.size})
     rng = np.random.default_rng(42 + comm.rank)
     comm.barrier()
 

@@ -45,6 +45,11 @@ class ArrowMatrix(ABC):
     def allgather_result(self, C):
         """All-gathers the result into C (numpy) and returns it."""
 
+    def set_features_slice_from_features(self, X):
+        """Deprecated in the reference (arrow_matrix.py:84-92): set this
+        rank's slice from the full feature matrix."""
+        raise NotImplementedError
+
     @staticmethod
     def column_subgroup(tiles_per_side, group):
         return group

@@ -363,6 +363,15 @@ class ArrowSlimMPI(ArrowMatrix):
     def feature_tile(self):
         return self.X_i
 
+    def set_features_slice_from_features(self, X) -> None:
+        """Deprecated in the reference (arrow_slim_mpi.py:434-440) but part
+        of the kept ABC: set this rank's stripe from the FULL feature
+        matrix."""
+        w = self.width
+        assert X.shape[0] % self.tiles_per_side == 0 or \
+            X.shape[0] >= self.last_block * w
+        self.set_features(X[self.first_block * w: self.last_block * w])
+
     def result_tile(self):
         return self.C_i
 
