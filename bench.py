@@ -209,10 +209,14 @@ def main():
 
     n_blocks = np.full(L, nb, dtype=np.int64)
     if L > 1:
-        # identity permutations between parts -> local-heavy exchange, but the
-        # full alltoallv path runs
-        to_prev = [None] + [np.arange(nb * w, dtype=np.int64)[first * w:last * w]] * (L - 1)
-        to_next = [np.arange(nb * w, dtype=np.int64)[first * w:last * w]] * (L - 1) + [None]
+        # a REAL random permutation between parts (deterministic) so the
+        # forward/backward exchange routes rows across all ranks
+        # (the alltoallv machinery is exercised at L >= 2)
+        rng_p = np.random.default_rng(1234)
+        perm = rng_p.permutation(nb * w).astype(np.int64)
+        inv = np.argsort(perm)
+        to_prev = [None] + [inv[first * w:last * w]] * (L - 1)
+        to_next = [perm[first * w:last * w]] * (L - 1) + [None]
     else:
         to_prev, to_next = [None], [None]
 
@@ -315,13 +319,22 @@ def main():
             total_bytes += 8.0 * nnz + 4.0 * (c_rows + 1) + 4.0 * k * (x_rows + c_rows)
         achieved = total_bytes / (total_ms / 1e3) / 1e9  # GB/s
         peak = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+        # measured per-launch PMC traffic, if tools/measure_traffic.py ran
+        # for this workload (ARROW_TRAFFIC_JSON); else null
+        traffic = None
+        tj = os.environ.get('ARROW_TRAFFIC_JSON')
+        if tj and os.path.exists(tj):
+            with open(tj) as f:
+                t = json.load(f)
+            if t.get('workload', {}).get('rows') == args.rows:
+                traffic = round(t['avg_read_bytes'] + t['avg_write_bytes_raw'])
         roofline = {
             "bound": "hbm",
             "achieved": round(achieved, 1),
             "peak": peak,
             "unit": "GB/s",
             "frac": round(achieved / peak, 4),
-            "traffic": None,
+            "traffic": traffic,
             "kernel": "spmm_kernel",
             "launches": len(events),
             "avg_launch_ms": round(total_ms / len(events), 4),
