@@ -164,27 +164,34 @@ def main():
     # loaded+scattered by the REFERENCE loader with a recording FakeComm.
     import tempfile
     loader_cases = [
-        dict(width=6, n_blocks=[3], seed=11, one_based=False),
-        dict(width=5, n_blocks=[4, 2], seed=12, one_based=False),
-        dict(width=4, n_blocks=[4, 3, 2], seed=13, one_based=False),
-        dict(width=6, n_blocks=[2, 2], seed=14, one_based=True),
+        dict(width=6, n_blocks=[3], seed=11, one_based=False, banded=False),
+        dict(width=5, n_blocks=[4, 2], seed=12, one_based=False, banded=False),
+        dict(width=4, n_blocks=[4, 3, 2], seed=13, one_based=False, banded=False),
+        dict(width=6, n_blocks=[2, 2], seed=14, one_based=True, banded=False),
+        dict(width=5, n_blocks=[4], seed=15, one_based=False, banded=True),
+        dict(width=4, n_blocks=[3, 3], seed=16, one_based=False, banded=True),
     ]
     for ci, cfg in enumerate(loader_cases):
+        banded = cfg.get('banded', False)
         decomp = synth.synth_arrow_decomposition(cfg['width'], cfg['n_blocks'],
-                                                 avg_deg=4, seed=cfg['seed'])
+                                                 avg_deg=4, seed=cfg['seed'],
+                                                 block_diagonal=not banded)
         if cfg['one_based']:
             decomp = [(B, p + 1) for B, p in decomp]
         with tempfile.TemporaryDirectory() as td:
             prefix = os.path.join(td, 'g')
-            my_graphio.save_decomposition_new(decomp, prefix, cfg['width'])
-            total = sum(cfg['n_blocks'])
+            my_graphio.save_decomposition_new(decomp, prefix, cfg['width'],
+                                              block_diagonal=not banded)
+            total = (sum(cfg['n_blocks']) if not banded
+                     else sum(2 * b - 1 for b in cfg['n_blocks']))
             comm = FakeComm(total)
             blocks0, n_blocks, to_prev0, to_next0 = ADM.load_decomposition_new(
-                comm, prefix, cfg['width'], is_block_diagonal=True, slim=True,
-                use_npy=True, use_mmap=False)
+                comm, prefix, cfg['width'], is_block_diagonal=not banded,
+                slim=not banded, use_npy=True, use_mmap=False)
         out[f'load_{ci}_n_blocks'] = np.asarray(n_blocks)
         out[f'load_{ci}_meta'] = np.asarray([cfg['width'], len(cfg['n_blocks']),
-                                             int(cfg['one_based']), cfg['seed']])
+                                             int(cfg['one_based']), cfg['seed'],
+                                             int(banded)])
         if to_next0 is not None:
             out[f'load_{ci}_rank0_to_next'] = to_next0
         # rank-0's own blocks (row 0 of part 0: A_00 only in slim scatter)

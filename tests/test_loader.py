@@ -70,22 +70,30 @@ def _assert_blocks_equal(a, b, msg=''):
 def test_loader_matches_reference_scatter(golden):
     n_cases = int(golden['load_n_cases'][0])
     for ci in range(n_cases):
-        width, L, one_based, seed = (int(x) for x in golden[f'load_{ci}_meta'])
+        meta = [int(x) for x in golden[f'load_{ci}_meta']]
+        width, L, one_based, seed = meta[:4]
+        banded = bool(meta[4]) if len(meta) > 4 else False
         n_blocks_ref = golden[f'load_{ci}_n_blocks']
         # regenerate the exact same decomposition + files
         nb_list = list(n_blocks_ref)
-        decomp = synth.synth_arrow_decomposition(width, nb_list, avg_deg=4, seed=seed)
+        decomp = synth.synth_arrow_decomposition(width, nb_list, avg_deg=4,
+                                                 seed=seed,
+                                                 block_diagonal=not banded)
         if one_based:
             decomp = [(B, p + 1) for B, p in decomp]
         with tempfile.TemporaryDirectory() as td:
             prefix = os.path.join(td, 'g')
-            graphio.save_decomposition_new(decomp, prefix, width)
-            P = int(sum(nb_list))
+            graphio.save_decomposition_new(decomp, prefix, width,
+                                           block_diagonal=not banded)
+            # my shared-rank loader: P ranks = max blocks so that each rank
+            # owns exactly one block-row of every part
+            P = int(max(nb_list))
             per_rank = []
             for r in range(P):
                 blocks, n_blocks, to_prev, to_next = \
                     ArrowDecompositionMPI.load_decomposition_new(
-                        _RankComm(r, P), prefix, width, is_block_diagonal=True)
+                        _RankComm(r, P), prefix, width,
+                        is_block_diagonal=not banded)
                 np.testing.assert_array_equal(n_blocks, n_blocks_ref)
                 per_rank.append((blocks, to_prev, to_next))
 
@@ -98,11 +106,14 @@ def test_loader_matches_reference_scatter(golden):
         if f'load_{ci}_rank0_to_next' in golden:
             np.testing.assert_array_equal(my_tn0[0], golden[f'load_{ci}_rank0_to_next'])
 
-        # every other (matrix i, block r) -> reference dest base_i + r
         base = 0
         for i, nb in enumerate(nb_list):
+            # reference rank layout: slim -> nb ranks per part (dest base+r);
+            # non-slim -> 2nb-1 ranks (row-0 columns at base+c, block-row r
+            # at base+nb-1+r, arrow_dec_mpi.py:753-823)
             for r in range(nb):
-                dest = base + r
+                dest = base + r if not banded else (base + nb - 1 + r if r > 0
+                                                    else base)
                 if dest == 0:
                     continue
                 blocks_d, perms_d = decoded[dest]
@@ -111,6 +122,12 @@ def test_loader_matches_reference_scatter(golden):
                 if r == 0:
                     # reference base rank of part i>0 receives A_00 (+ perms)
                     exp = [grid[0][0]]
+                elif banded:
+                    exp = [grid[r][0], grid[r][r]]
+                    if r > 1:
+                        exp.append(grid[r][r - 1])
+                    if r < nb - 1:
+                        exp.append(grid[r][r + 1])
                 else:
                     exp = [grid[0][r], grid[r][0], grid[r][r]]
                 assert len(blocks_d) == len(exp), f"dest {dest}: block count"
@@ -126,4 +143,13 @@ def test_loader_matches_reference_scatter(golden):
                 assert len(perms_d) == len(expected_perms), f"dest {dest}: perm count"
                 for got, mine in zip(perms_d, expected_perms):
                     np.testing.assert_array_equal(got, mine)
-            base += nb
+            if banded:
+                # row-0 column tiles at base+1..base+nb-1 (A_0c only)
+                for c in range(1, nb):
+                    blocks_d, perms_d = decoded[base + c]
+                    grid = per_rank[c][0][i]
+                    assert len(blocks_d) == 1, f"col dest {base+c}"
+                    _assert_blocks_equal(blocks_d[0].toarray(),
+                                         grid[0][c].toarray(),
+                                         f'part {i} A_0{c}')
+            base += nb if not banded else 2 * nb - 1
