@@ -156,6 +156,23 @@ def save_decomposition(decomposition, filename: str, width: int,
                 np.asarray(permutation))
 
 
+def convert_decomposition(filename: str, width: Optional[int] = None,
+                          block_diagonal: bool = True):
+    """Convert a legacy .npz decomposition to the .npy CSR files
+    (reference graphio.py:317-358)."""
+    decomposition = load_decomposition(filename, width, block_diagonal,
+                                       no_permutation=True)
+    for i, (B, _) in enumerate(decomposition):
+        B = sparse.csr_matrix(B)
+        np.save(format_path(filename, width, i, block_diagonal,
+                            DecompositionFileType.indptr_npy), B.indptr)
+        np.save(format_path(filename, width, i, block_diagonal,
+                            DecompositionFileType.indices_npy), B.indices)
+        np.save(format_path(filename, width, i, block_diagonal,
+                            DecompositionFileType.data_npy), B.data)
+    return decomposition
+
+
 def split_matrix_to_blocks(A: sparse.csr_matrix, block_size: int,
                            dtype=None, use_min_shape: bool = False
                            ) -> List[List[Union[sparse.csr_matrix, None]]]:
