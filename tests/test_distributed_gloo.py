@@ -11,7 +11,6 @@ import pytest
 import torch
 import torch.multiprocessing as mp
 
-WORLD = 2
 
 
 def _free_port():
@@ -22,11 +21,11 @@ def _free_port():
     return p
 
 
-def _worker(rank, port, tmpdir, result_q):
+def _worker(rank, port, tmpdir, result_q, world=2):
     import torch.distributed as dist
     os.environ['MASTER_ADDR'] = '127.0.0.1'
     os.environ['MASTER_PORT'] = str(port)
-    dist.init_process_group('gloo', rank=rank, world_size=WORLD)
+    dist.init_process_group('gloo', rank=rank, world_size=world)
     try:
         from arrow_matrix_amd import graphio, synth
         from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
@@ -92,13 +91,16 @@ def _worker(rank, port, tmpdir, result_q):
         dist.destroy_process_group()
 
 
-def test_distributed_gloo_world2():
+@pytest.mark.parametrize("world", [2, 3])
+def test_distributed_gloo(world):
+    # world=3 exercises uneven spans including a rank owning NO blocks of
+    # small parts (nb=2 over 3 ranks)
     port = _free_port()
     ctx = mp.get_context('spawn')
     q = ctx.Queue()
     with tempfile.TemporaryDirectory() as td:
-        procs = [ctx.Process(target=_worker, args=(r, port, td, q))
-                 for r in range(WORLD)]
+        procs = [ctx.Process(target=_worker, args=(r, port, td, q, world))
+                 for r in range(world)]
         for p in procs:
             p.start()
         for p in procs:

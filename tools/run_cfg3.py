@@ -1,0 +1,128 @@
+#!/usr/bin/env python3
+"""BASELINE configs[2] (cfg3) end-to-end run: roadNet-CA is not shippable
+(no network), so — as SURVEY.md §8d prescribes — a 2M-row planar-like
+synthetic substitute is used and STATED: ring + short-range edges + a
+zipf-skewed long-range tail, ~5.5M nnz, symmetric. width=500k, k=32.
+
+Flow: numpy-native arrow_decompose -> .npy files -> loader -> GPU engine;
+parity vs scipy A @ X at FULL size; then a timed iterated-SpMM loop.
+Writes gpurun_out/cfg3.json.
+"""
+import json
+import os
+import sys
+import time
+
+import numpy as np
+from scipy import sparse
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+
+def planar_like(n, seed):
+    rng = np.random.default_rng(seed)
+    src = [np.arange(n - 1)]
+    dst = [np.arange(1, n)]                      # ring/backbone
+    m_near = 2 * n
+    s = rng.integers(0, n, m_near)
+    d = (s + rng.integers(1, 64, m_near)) % n     # short-range
+    src.append(s); dst.append(d)
+    m_far = n // 4
+    s = rng.integers(0, n, m_far)
+    u = rng.random(m_far)
+    d = (u * u * u * u * n).astype(np.int64).clip(0, n - 1)  # zipf tail -> hubs
+    src.append(s); dst.append(d)
+    src = np.concatenate(src); dst = np.concatenate(dst)
+    keep = src != dst
+    A = sparse.csr_matrix((np.ones(keep.sum(), np.float32),
+                           (src[keep], dst[keep])), shape=(n, n))
+    A = sparse.csr_matrix(A.maximum(A.T))
+    A.data[:] = 1.0
+    return A
+
+
+def main():
+    import torch
+    from arrow_matrix_amd import graphio
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+    from arrow_matrix_amd.decomposition import arrow_decomposition
+
+    n, width, k = 2_000_000, 500_000, 32
+    t0 = time.perf_counter()
+    A = planar_like(n, 42)
+    print(f"# graph: n={n} nnz={A.nnz} ({time.perf_counter()-t0:.1f}s)",
+          file=sys.stderr)
+
+    t0 = time.perf_counter()
+    decomp = arrow_decomposition(A, arrow_width=width,
+                                 max_number_of_levels=10, block_diagonal=True,
+                                 rng=np.random.default_rng(0))
+    print(f"# decomposed into {len(decomp)} parts "
+          f"({time.perf_counter()-t0:.1f}s)", file=sys.stderr)
+
+    out_dir = os.path.join(REPO, 'gpurun_out')
+    os.makedirs(out_dir, exist_ok=True)
+    prefix = os.path.join('/tmp', 'cfg3')
+    pairs = [(p.graph, p.permutation) for p in decomp]
+    graphio.save_decomposition_new(pairs, prefix, width)
+
+    blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+        None, prefix, width)
+    arrow = ArrowDecompositionMPI.initialize(None, nb, tp, tn, width, k,
+                                             device='gpu')
+    arrow.load_data_from_blocks(blocks)
+    arrow.zero_rhs(width, k)
+
+    rng = np.random.default_rng(7)
+    X = (2 * rng.random((n, k), dtype=np.float32) - 1)
+    perm0 = pairs[0][1]
+    n_pad = int(nb[0]) * width
+    X_eng = np.zeros((n_pad, k), np.float32)
+    m = min(n, n_pad)
+    X_eng[:m] = X[perm0][:m]
+    arrow.B.set_features(X_eng.copy())
+
+    # parity at full size vs scipy A @ X (the reference's own golden)
+    arrow.step()
+    C = arrow.B.allgather_result()
+    golden = (A @ X)[perm0]
+    err = np.abs(C[:m] - golden[:m]).max()
+    scale = max(1.0, np.abs(golden).max())
+    print(f"# parity max |err| = {err:.3e} (scale {scale:.1f})", file=sys.stderr)
+    assert err <= 1e-4 * scale, "cfg3 parity failed"
+
+    # timed iterated loop
+    for eng in arrow.engines:
+        eng.allreduce_x0 = True
+    import torch
+    steps, warmup = 20, 3
+    for _ in range(warmup):
+        arrow.step()
+        arrow.B.set_features(arrow.B.result_tile())
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        arrow.step()
+        arrow.B.set_features(arrow.B.result_tile())
+    torch.cuda.synchronize()
+    el = time.perf_counter() - t0
+    nnz = sum(p.graph.nnz for p in decomp)
+    res = {
+        "config": "cfg3 (2M-row planar-like synthetic substitute for "
+                  "roadNet-CA, stated per SURVEY.md §8d)",
+        "rows": n, "nnz": int(nnz), "width": width, "features": k,
+        "parts": len(decomp), "n_blocks": [int(x) for x in nb],
+        "parity_max_err": float(err),
+        "steps": steps, "ms_per_step": round(el / steps * 1e3, 3),
+        "gflops": round(2.0 * nnz * k * steps / el / 1e9, 2),
+        "n_gpus": 1,
+    }
+    line = json.dumps(res)
+    print(line)
+    with open(os.path.join(out_dir, 'cfg3.json'), 'w') as f:
+        f.write(line + "\n")
+
+
+if __name__ == '__main__':
+    main()
