@@ -183,6 +183,10 @@ class ArrowSlimMPI(ArrowMatrix):
                     and self.comm.size == 1 and self.first_block == 0
                     and self.n_owned == self.tiles_per_side)
         # --- row-0 merge: C_0 = [A_0,first .. A_0,last-1] @ X_stripe -------
+        # Built in ROW CHUNKS so each chunk's C_0 slice can start its
+        # (all)reduce while later chunks and the rest launch still compute:
+        # at N=8 the 6.4 GB collective pipelines behind compute instead of
+        # serialising after one monolithic launch.
         rows_cat, cols_cat, data_cat = [], [], []
         for j, b in enumerate(self.A_0i):
             b = b.tocsr()
@@ -196,8 +200,19 @@ class ArrowSlimMPI(ArrowMatrix):
             # X_0 == X_i[:w] needs no broadcast copy
             row0_sets = (rows_cat, cols_cat, data_cat)
         else:
-            self._A_row0 = self._merged_handle(w, nw * w, rows_cat, cols_cat,
-                                               data_cat, x_rows=nw * w)
+            n_chunks = 4 if (w >= 64 and self.comm.size > 1) else 1
+            rows = np.concatenate(rows_cat)
+            cols = np.concatenate(cols_cat)
+            data = np.concatenate(data_cat)
+            bounds = [w * q // n_chunks for q in range(n_chunks + 1)]
+            self._A_row0 = []
+            for q in range(n_chunks):
+                lo, hi = bounds[q], bounds[q + 1]
+                m = (rows >= lo) & (rows < hi)
+                h = self._merged_handle(hi - lo, nw * w,
+                                        [rows[m] - lo], [cols[m]], [data[m]],
+                                        x_rows=nw * w // n_chunks)
+                self._A_row0.append((h, lo, hi))
         # --- rest merge: C[r] = A_rr @ X_r + A_r0 @ X_0
         #     (+ interior banded off-diagonals A_{r,r±1} @ X_{r±1} when the
         #      neighbour block is owned; boundary off-diagonals become small
@@ -456,24 +471,33 @@ class ArrowSlimMPI(ArrowMatrix):
                         h.nnz, self.n_owned * w, h.x_rows)
             return
 
-        # C_0 = A_row0_merged @ X_stripe  (ONE launch; the reference runs one
-        # CSRMM per block and re-uploads A and X, arrow_slim_mpi.py:181-195)
-        if self._A_row0 is not None:
-            h = self._A_row0
-            self._timed(lambda: be.spmm_block(h, self.X_i, self.C_0, 0),
-                        h.nnz, w, h.x_rows)
+        # C_0 = A_row0_merged @ X_stripe in row chunks (the reference runs
+        # one CSRMM per block and re-uploads A and X,
+        # arrow_slim_mpi.py:181-195); each chunk's (all)reduce starts as
+        # soon as its slice is computed and pipelines behind the remaining
+        # compute (arrow_slim_mpi.py:116's single blocking Reduce).
+        tic = time.perf_counter()
+        reduce_works = []
+        if self._A_row0:
+            for h, lo, hi in self._A_row0:
+                C_sl = self.C_0[lo:hi]
+                self._timed(lambda h=h, C_sl=C_sl:
+                            be.spmm_block(h, self.X_i, C_sl, 0),
+                            h.nnz, hi - lo, h.x_rows)
+                if self.allreduce_x0:
+                    reduce_works.append(
+                        self.comm.allreduce_sum_(C_sl, async_op=True))
+                else:
+                    reduce_works.append(
+                        self.comm.reduce_sum_(C_sl, dst=0, async_op=True))
         else:
             self.C_0.zero_()
-
-        # ONE reduce of the first block-row partials (arrow_slim_mpi.py:116),
-        # overlapped with the rest launch below. In the iterated loop an
-        # ALLREDUCE leaves the reduced C_0 on every rank — it becomes the
-        # next iteration's X_0 without a broadcast.
-        tic = time.perf_counter()
-        if self.allreduce_x0:
-            reduce_work = self.comm.allreduce_sum_(self.C_0, async_op=True)
-        else:
-            reduce_work = self.comm.reduce_sum_(self.C_0, dst=0, async_op=True)
+            if self.allreduce_x0:
+                reduce_works.append(
+                    self.comm.allreduce_sum_(self.C_0, async_op=True))
+            else:
+                reduce_works.append(
+                    self.comm.reduce_sum_(self.C_0, dst=0, async_op=True))
         wb_logging.log({"spmm_row_reduce": time.perf_counter() - tic})
 
         # C_rest = A_diag_merged @ X_stripe + A_col_merged @ X_0 fused:
@@ -501,8 +525,9 @@ class ArrowSlimMPI(ArrowMatrix):
                 Cr = self.C_i[local:local + w]
                 self._timed(lambda: be.spmm_block(hdl, halo, Cr, 1),
                             hdl.nnz, w, w)
-        if reduce_work is not None:
-            reduce_work.wait()
+        for wk in reduce_works:
+            if wk is not None:
+                wk.wait()
 
     def _exchange_halos(self) -> None:
         """±1 halo X exchange for the banded mode: my first owned tile goes

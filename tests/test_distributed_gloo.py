@@ -40,6 +40,7 @@ def _worker(rank, port, tmpdir, result_q, world=2):
             ([4, 3, 2], 4, 6, 3, False), # L=3 cascade
             ([4], 6, 4, 7, True),        # banded: cross-rank halo exchange
             ([5], 4, 3, 8, True),        # banded, uneven spans (3+2)
+            ([2], 64, 4, 9, False),      # wide blocks -> chunked C_0 reduce
         ]
         for n_blocks, width, k, seed, banded in cases:
             decomp = synth.synth_arrow_decomposition(width, n_blocks,

@@ -86,8 +86,11 @@ def test_merged_structures_match_per_block(nb, w, first, last):
                                        rtol=1e-5, atol=1e-6)
         return
 
-    # row-0 merged: C_0 = sum_c A_0c @ X_c
-    C0 = _apply_merged(eng._A_row0, X_stripe, X_0)
+    # row-0 merged: C_0 = sum_c A_0c @ X_c (list of row chunks; single
+    # chunk at these sizes/world=1)
+    C0 = np.zeros_like(C0_ref)
+    for h, lo, hi in eng._A_row0:
+        C0[lo:hi] = _apply_merged(h, X_stripe, X_0)
     np.testing.assert_allclose(C0, C0_ref, rtol=1e-5, atol=1e-6)
 
     # rest merged: C_r = A_rr @ X_r + A_r0 @ X_0
