@@ -201,6 +201,9 @@ class ArrowSlimMPI(ArrowMatrix):
             row0_sets = (rows_cat, cols_cat, data_cat)
         else:
             n_chunks = 4 if (w >= 64 and self.comm.size > 1) else 1
+            env_c = _os.environ.get('ARROW_ROW0_CHUNKS')
+            if env_c:
+                n_chunks = max(1, min(int(env_c), w))
             rows = np.concatenate(rows_cat)
             cols = np.concatenate(cols_cat)
             data = np.concatenate(data_cat)

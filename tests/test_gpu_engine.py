@@ -126,3 +126,20 @@ def test_engine_gpu_split_col(monkeypatch):
                                    device='gpu', seed=17)
     for C, G in zip(results, goldens):
         np.testing.assert_allclose(C, G, rtol=2e-4, atol=2e-4)
+
+
+@pytest.mark.gpu
+def test_engine_gpu_chunked_row0(monkeypatch):
+    """Force the chunked C_0 pipeline (ARROW_ROW0_CHUNKS=4) at world=1 so
+    its numerics are validated before the driver's multi-GPU run."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    monkeypatch.setenv('ARROW_ROW0_CHUNKS', '4')
+    from tests.test_engine_cpu import _run_engine
+    from arrow_matrix_amd import synth
+    decomp = synth.synth_arrow_decomposition(64, [4], avg_deg=8, seed=23,
+                                             hub_rows=2, hub_deg=100)
+    results, goldens = _run_engine(decomp, 64, [4], 16, iters=2,
+                                   device='gpu', seed=23)
+    for C, G in zip(results, goldens):
+        np.testing.assert_allclose(C, G, rtol=2e-4, atol=2e-4)
