@@ -181,6 +181,9 @@ class ArrowDecompositionMPI:
                 recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
                 # C_i[recv_perm] = recvbuf; X := C (arrow_dec_mpi.py:544-545)
                 eng_r.backend.scatter_rows(eng_r.C_i, ex.recv_rows, recvbuf)
+            # the exchange mutated C in place: any cached X_0 (the
+            # allreduce_x0 fast path) is stale now
+            eng_r._x0_valid = False
             eng_r.set_features(eng_r.C_i)
 
     def _aggregate(self) -> None:
@@ -197,6 +200,7 @@ class ArrowDecompositionMPI:
                 recvbuf = self.comm.alltoallv(sendbuf, ex.send_counts, ex.recv_counts)
                 # C_i[recv_perm] += recvbuf; X := C (arrow_dec_mpi.py:437-438)
                 eng_r.backend.scatter_add_rows(eng_r.C_i, ex.recv_rows, recvbuf)
+            eng_r._x0_valid = False  # C mutated after spmm; cached X_0 stale
             eng_r.set_features(eng_r.C_i)
 
     # -- loading -------------------------------------------------------------

@@ -233,8 +233,10 @@ def main():
     eng0 = arrow.engines[0]
     for eng in arrow.engines:
         # iterated GNN-propagation loop (X := C): fuse C_0 reduce with the
-        # next X_0 broadcast into one allreduce
-        eng.allreduce_x0 = (args.device == 'gpu')
+        # next X_0 broadcast into one allreduce. Only
+        # valid at L == 1: with multiple parts the inter-part exchange
+        # rewrites C after the spmm, so next X_0 != this C_0.
+        eng.allreduce_x0 = (args.device == 'gpu' and L == 1)
     if use_gpu:
         g = torch.Generator(device='cuda')
         g.manual_seed(42 + comm.rank)
