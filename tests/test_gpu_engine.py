@@ -143,3 +143,43 @@ def test_engine_gpu_chunked_row0(monkeypatch):
                                    device='gpu', seed=23)
     for C, G in zip(results, goldens):
         np.testing.assert_allclose(C, G, rtol=2e-4, atol=2e-4)
+
+
+@pytest.mark.gpu
+def test_allreduce_x0_multi_part_invalidation():
+    """With allreduce_x0 force-enabled at L=2, the exchange must invalidate
+    the cached X_0 (stale-X_0 regression test)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import os
+    import tempfile
+    from arrow_matrix_amd import graphio, synth
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+    from oracle import compute_spmm
+
+    n_blocks, width, k = [3, 2], 40, 8
+    decomp = synth.synth_arrow_decomposition(width, n_blocks, avg_deg=5, seed=41)
+    n = n_blocks[0] * width
+    rng = np.random.default_rng(41)
+    X_orig = (2 * rng.random((n, k)) - 1).astype(np.float32)
+    with tempfile.TemporaryDirectory() as td:
+        prefix = os.path.join(td, 'g')
+        graphio.save_decomposition_new(decomp, prefix, width)
+        blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+            None, prefix, width)
+        arrow = ArrowDecompositionMPI.initialize(None, nb, tp, tn, width, k,
+                                                 device='gpu')
+        arrow.load_data_from_blocks(blocks)
+        arrow.zero_rhs(width, k)
+        for eng in arrow.engines:
+            eng.allreduce_x0 = True  # deliberately forced at L>1
+        perm0 = decomp[0][1]
+        arrow.B.set_features(X_orig[perm0].copy())
+        golden_X = X_orig
+        for _ in range(3):
+            arrow.step()
+            C = arrow.B.allgather_result()
+            golden = compute_spmm(decomp, golden_X)[perm0]
+            np.testing.assert_allclose(C, golden, rtol=5e-4, atol=5e-4)
+            golden_X = compute_spmm(decomp, golden_X)
+            arrow.B.set_features(arrow.B.result_tile())
