@@ -13,6 +13,8 @@ pytestmark = pytest.mark.gpu
     ([3], 64, 32, 1),
     ([4, 2], 50, 8, 2),
     ([4, 3, 2], 32, 16, 3),
+    ([3], 48, 300, 4),         # k > 256: column-tiled dual-operand launches
+    ([2], 40, 7, 5),           # odd k (VEC=1 path) through the fused engine
 ])
 def test_engine_gpu_matches_golden(n_blocks, width, k, seed):
     if not torch.cuda.is_available():
