@@ -86,7 +86,8 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
     const int32_t *__restrict__ item_row, const int32_t *__restrict__ item_begin,
     const int32_t *__restrict__ item_end, int64_t n_items,
     const float *__restrict__ X0, const float *__restrict__ X1,
-    float *__restrict__ C, int64_t k, int64_t col_off, int xcd_remap) {
+    float *__restrict__ C, int64_t k, int64_t col_off, int xcd_remap,
+    int nt_mode) {
   constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
   const int lane_in_group = threadIdx.x % GROUP;
   const int group_in_block = threadIdx.x / GROUP;
@@ -114,9 +115,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
   // dominates short power-law rows
   int32_t next_row = 0, next_b = 0, next_e = 0;
   if (item < n_items) {
-    next_row = item_row[item];
-    next_b = item_begin[item];
-    next_e = item_end[item];
+    next_row = __builtin_nontemporal_load(&item_row[item]);
+    next_b = __builtin_nontemporal_load(&item_begin[item]);
+    next_e = __builtin_nontemporal_load(&item_end[item]);
   }
 
   for (; item < n_items; item += stride) {
@@ -127,9 +128,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
     const int32_t e = next_e;
     const int64_t nxt = item + stride;
     if (nxt < n_items) {
-      next_row = item_row[nxt];
-      next_b = item_begin[nxt];
-      next_e = item_end[nxt];
+      next_row = __builtin_nontemporal_load(&item_row[nxt]);
+      next_b = __builtin_nontemporal_load(&item_begin[nxt]);
+      next_e = __builtin_nontemporal_load(&item_end[nxt]);
     }
 
     float acc[VEC];
@@ -229,7 +230,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
     };
     int32_t base = b;
     for (; base + GROUP <= e; base += GROUP) {  // full chunks
-      const int2 mine = pairs[base + lane_in_group];
+      const int2 mine = nt_mode
+          ? __builtin_nontemporal_load(&pairs[base + lane_in_group])
+          : pairs[base + lane_in_group];
       if constexpr (GROUP >= 8) {
 #pragma unroll
         for (int u = 0; u < GROUP; u += 8) consume8(mine, u);
@@ -270,7 +273,11 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
       } else {
         if constexpr (VEC == 4) {
           float4 outv{acc[0], acc[1], acc[2], acc[3]};
-          *reinterpret_cast<float4 *>(cr) = outv;
+          if (nt_mode) {
+            __builtin_nontemporal_store(outv, reinterpret_cast<float4 *>(cr));
+          } else {
+            *reinterpret_cast<float4 *>(cr) = outv;
+          }
         } else {
 #pragma unroll
           for (int j = 0; j < VEC; ++j) cr[j] = acc[j];
@@ -444,7 +451,7 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
       hipLaunchKernelGGL(kern, dim3(blocks), dim3(BLOCK_THREADS), 0, stream,
                          blk.pairs, blk.item_row, blk.item_begin,
                          blk.item_end, blk.n_items, X0, X1, C, k, col_off,
-                         blk.xcd_remap);
+                         blk.xcd_remap, nt_mode);
     };
     if (beta == 0) {
       if (guard) run(spmm_kernel<VEC, GROUP, 0, true>);
