@@ -63,3 +63,43 @@ def test_fuzz_case(case):
                        for B, p in decomp]
         golden = compute_spmm(zero_decomp, X)[perm0]
         np.testing.assert_allclose(C, golden, rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("case", CASES[:8])
+def test_fuzz_case_gpu(case):
+    """The same randomised sweep through the GPU engine."""
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    blocked = not case['banded']
+    decomp = synth.synth_arrow_decomposition(
+        case['width'], case['n_blocks'], avg_deg=int(np.random.default_rng(
+            case['seed']).integers(2, 8)), seed=case['seed'],
+        block_diagonal=blocked, hub_rows=case['hub_rows'])
+    if case['one_based']:
+        decomp = [(B, p + 1) for B, p in decomp]
+    n = case['n_blocks'][0] * case['width']
+    rng = np.random.default_rng(case['seed'] + 1)
+    X = (2 * rng.random((n, case['k'])) - 1).astype(np.float32)
+    with tempfile.TemporaryDirectory() as td:
+        prefix = os.path.join(td, 'g')
+        graphio.save_decomposition_new(decomp, prefix, case['width'],
+                                       block_diagonal=blocked)
+        blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+            None, prefix, case['width'], is_block_diagonal=blocked)
+        arrow = ArrowDecompositionMPI.initialize(
+            None, nb, tp, tn, case['width'], case['k'], device='gpu',
+            block_diagonal=blocked, slim=blocked)
+        arrow.load_data_from_blocks(blocks)
+        arrow.zero_rhs(case['width'], case['k'])
+        perm0 = np.asarray(decomp[0][1], dtype=np.int64)
+        if case['one_based']:
+            perm0 = perm0 - 1
+        arrow.B.set_features(X[perm0].copy())
+        arrow.step()
+        C = arrow.B.allgather_result()
+        zero_decomp = [(B, (p - 1 if case['one_based'] else p))
+                       for B, p in decomp]
+        golden = compute_spmm(zero_decomp, X)[perm0]
+        np.testing.assert_allclose(C, golden, rtol=2e-4, atol=2e-4)
