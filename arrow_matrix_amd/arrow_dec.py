@@ -145,9 +145,9 @@ class ArrowDecompositionMPI:
         for eng, blocks in zip(self.engines, blocked):
             eng.load_sparse_matrix_from_blocks(blocks)
 
-    def zero_rhs(self, width: int, n_features: int) -> None:
+    def zero_rhs(self, width: int, n_features: int, dtype=np.float32) -> None:
         for eng in self.engines:
-            eng.zero_rhs(width, n_features)
+            eng.zero_rhs(width, n_features, dtype=dtype)
 
     # -- iteration -----------------------------------------------------------
 

@@ -334,6 +334,15 @@ class ArrowSlimMPI(ArrowMatrix):
     def zero_rhs(self, number_of_rows_per_rank: int, number_of_columns: int,
                  dtype=np.float32) -> None:
         assert number_of_rows_per_rank >= 1 and number_of_columns >= 1
+        if np.dtype(dtype) != np.float32:
+            # the reference chooses the dtype at zero_rhs time
+            # (arrow_slim_mpi.py:354)
+            if self.backend.device == 'cuda':
+                raise NotImplementedError(
+                    "float64 runs on device='cpu' (HIP kernels are fp32)")
+            if np.dtype(dtype) != self.backend.np_dtype:
+                self.backend = make_backend(self.device, dtype)
+                self.C_i = self.C_0 = self.X_i = self.X_0 = None
         self.width = number_of_rows_per_rank
         w, k = number_of_rows_per_rank, number_of_columns
         stripe = (max(self.n_owned, 1) * w, k)

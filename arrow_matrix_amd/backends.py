@@ -18,16 +18,24 @@ import torch
 from . import hip
 
 
+_TORCH_DTYPE = {np.dtype(np.float32): torch.float32,
+                np.dtype(np.float64): torch.float64}
+
+
 class CpuBackend:
     device = 'cpu'
 
+    def __init__(self, dtype=np.float32):
+        self.np_dtype = np.dtype(dtype)
+        self.torch_dtype = _TORCH_DTYPE[self.np_dtype]
+
     def zeros(self, shape):
-        return torch.zeros(shape, dtype=torch.float32)
+        return torch.zeros(shape, dtype=self.torch_dtype)
 
     def asarray(self, x) -> torch.Tensor:
         if isinstance(x, torch.Tensor):
-            return x.float().cpu()
-        return torch.from_numpy(np.ascontiguousarray(x, dtype=np.float32))
+            return x.to(self.torch_dtype).cpu()
+        return torch.from_numpy(np.ascontiguousarray(x, dtype=self.np_dtype))
 
     def index_tensor(self, idx: np.ndarray) -> torch.Tensor:
         return torch.from_numpy(np.ascontiguousarray(idx, dtype=np.int64))
@@ -142,9 +150,13 @@ class GpuBackend:
         torch.cuda.synchronize(self.device_index)
 
 
-def make_backend(device: str):
+def make_backend(device: str, dtype=np.float32):
     if device in ('gpu', 'cuda'):
+        if np.dtype(dtype) != np.float32:
+            raise NotImplementedError(
+                "the HIP kernels compute in fp32 (the reference benchmark "
+                "default, arrow_bench.py:21); float64 runs on device='cpu'")
         return GpuBackend()
     if device == 'cpu':
-        return CpuBackend()
+        return CpuBackend(dtype)
     raise ValueError(f"unknown device {device!r} (use 'cpu' or 'gpu')")

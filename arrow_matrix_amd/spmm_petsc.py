@@ -31,10 +31,10 @@ class SpmmPETSc:
     """Resident 1D-slice SpMM engine over a MatrixSlice."""
 
     def __init__(self, comm: Optional[Comm], matrix_slice: MatrixSlice,
-                 device: str = 'cpu'):
+                 device: str = 'cpu', dtype=np.float32):
         self.comm = comm if comm is not None else Comm()
         self.ms = matrix_slice
-        self.backend = make_backend(device)
+        self.backend = make_backend(device, dtype)
         self.A_local = self.backend.upload_block(matrix_slice.A_i_local)
         self.A_nonlocal = (self.backend.upload_block(matrix_slice.A_i_nonlocal)
                            if matrix_slice.A_i_nonlocal.shape[1] > 0 else None)

@@ -5,6 +5,7 @@ import tempfile
 
 import numpy as np
 import pytest
+import torch
 
 from arrow_matrix_amd import graphio, synth
 from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
@@ -92,3 +93,46 @@ def test_gpu_device_without_gpu_fails_loudly():
     from arrow_matrix_amd.hip import ArrowSpmmError
     with pytest.raises((ArrowSpmmError, RuntimeError)):
         make_backend('gpu')
+
+
+def test_engine_cpu_float64():
+    """float64 on the cpu device (the reference's -t float64 option,
+    spmm_petsc_main.py / datatype param): full-precision scipy path."""
+    from arrow_matrix_amd import graphio
+    n_blocks, width, k = [3], 6, 4
+    decomp = synth.synth_arrow_decomposition(width, n_blocks, avg_deg=5, seed=77)
+    n = n_blocks[0] * width
+    rng = np.random.default_rng(77)
+    X = (2 * rng.random((n, k)) - 1)  # float64
+    with tempfile.TemporaryDirectory() as td:
+        prefix = os.path.join(td, 'g')
+        graphio.save_decomposition_new(decomp, prefix, width)
+        blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+            None, prefix, width, datatype=np.float64)
+        arrow = ArrowDecompositionMPI.initialize(None, nb, tp, tn, width, k,
+                                                 device='cpu')
+        arrow.load_data_from_blocks(blocks)
+        arrow.zero_rhs(width, k, dtype=np.float64)
+        perm0 = decomp[0][1]
+        arrow.B.set_features(X[perm0].copy())
+        arrow.step()
+        C = arrow.B.allgather_result()
+        assert C.dtype == np.float64
+        A = synth.recompose(decomp).astype(np.float64)
+        golden = (A @ X)[perm0]
+        np.testing.assert_allclose(C, golden, rtol=1e-12, atol=1e-12)
+
+
+def test_petsc_cpu_float64():
+    from scipy import sparse
+    from arrow_matrix_amd.matrix_slice import MatrixSlice
+    from arrow_matrix_amd.spmm_petsc import SpmmPETSc
+    rng = np.random.RandomState(3)
+    A = sparse.csr_matrix(sparse.random(20, 20, density=0.3, random_state=rng,
+                                        format='csr'), dtype=np.float64)
+    ms = MatrixSlice.initialize(None, A)
+    eng = SpmmPETSc(None, ms, device='cpu', dtype=np.float64)
+    X = rng.rand(20, 3)
+    Y = eng.spmm(X)
+    assert Y.dtype == torch.float64
+    np.testing.assert_allclose(Y.numpy(), A @ X, rtol=1e-12, atol=1e-12)
