@@ -165,6 +165,17 @@ class ArrowSlimMPI(ArrowMatrix):
                     blocks[j] = b[:, nnz_cols]
                     self._nnz_columns[j][slot] = nnz_cols
 
+    def _row0_chunks(self, w: int) -> int:
+        """Chunk count of the C_0 (all)reduce pipeline — a pure function of
+        (width, world size) so every rank issues the identical collective
+        schedule, including ranks that own no blocks of this matrix."""
+        import os as _os
+        n_chunks = 4 if (w >= 64 and self.comm.size > 1) else 1
+        env_c = _os.environ.get('ARROW_ROW0_CHUNKS')
+        if env_c:
+            n_chunks = max(1, min(int(env_c), w))
+        return n_chunks
+
     def _build_merged_gpu(self) -> None:
         w = self.width
         nw = self.n_owned
@@ -200,10 +211,7 @@ class ArrowSlimMPI(ArrowMatrix):
             # X_0 == X_i[:w] needs no broadcast copy
             row0_sets = (rows_cat, cols_cat, data_cat)
         else:
-            n_chunks = 4 if (w >= 64 and self.comm.size > 1) else 1
-            env_c = _os.environ.get('ARROW_ROW0_CHUNKS')
-            if env_c:
-                n_chunks = max(1, min(int(env_c), w))
+            n_chunks = self._row0_chunks(w)
             rows = np.concatenate(rows_cat)
             cols = np.concatenate(cols_cat)
             data = np.concatenate(data_cat)
@@ -503,13 +511,21 @@ class ArrowSlimMPI(ArrowMatrix):
                     reduce_works.append(
                         self.comm.reduce_sum_(C_sl, dst=0, async_op=True))
         else:
+            # a rank with no blocks of this matrix must still issue the SAME
+            # collective schedule as its peers (chunk count is a function of
+            # (width, world) only) — a single reduce here would deadlock
+            # against their chunked reduces
             self.C_0.zero_()
-            if self.allreduce_x0:
-                reduce_works.append(
-                    self.comm.allreduce_sum_(self.C_0, async_op=True))
-            else:
-                reduce_works.append(
-                    self.comm.reduce_sum_(self.C_0, dst=0, async_op=True))
+            n_chunks = self._row0_chunks(w)
+            bounds = [w * q // n_chunks for q in range(n_chunks + 1)]
+            for q in range(n_chunks):
+                C_sl = self.C_0[bounds[q]:bounds[q + 1]]
+                if self.allreduce_x0:
+                    reduce_works.append(
+                        self.comm.allreduce_sum_(C_sl, async_op=True))
+                else:
+                    reduce_works.append(
+                        self.comm.reduce_sum_(C_sl, dst=0, async_op=True))
         wb_logging.log({"spmm_row_reduce": time.perf_counter() - tic})
 
         # C_rest = A_diag_merged @ X_stripe + A_col_merged @ X_0 fused:
