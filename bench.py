@@ -184,9 +184,11 @@ def main():
     from arrow_matrix_amd.comm import TorchDistComm, Comm
 
     if world > 1:
-        backend = 'nccl' if use_gpu else 'gloo'
+        backend = os.environ.get('ARROW_BENCH_BACKEND',
+                                 'nccl' if use_gpu else 'gloo')
         if use_gpu:
-            torch.cuda.set_device(local_rank)
+            # modulo lets oversubscribed validation runs share one device
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
         comm = TorchDistComm()
     else:
