@@ -108,8 +108,13 @@ class TorchDistComm(Comm):
         Implemented as grouped point-to-point (batch_isend_irecv): the RCCL
         alltoallv idiom over xGMI, and the only form gloo also supports.
         The self-block is copied locally (never hits the backend).
+        CPU tensors under an nccl-only group (e.g. MatrixSlice's int
+        tables) are round-tripped through the device.
         """
         assert send.dim() == 2
+        if send.device.type == 'cpu' and dist.get_backend(self.group) == 'nccl':
+            recv = self.alltoallv(send.cuda(), send_counts, recv_counts)
+            return recv.cpu()
         sdispl = [0]
         for c in send_counts:
             sdispl.append(sdispl[-1] + int(c))
