@@ -162,6 +162,8 @@ class TorchDistComm(Comm):
         return recv, works
 
     def allgather_cat(self, tensor: torch.Tensor) -> torch.Tensor:
+        if tensor.device.type == 'cpu' and dist.get_backend(self.group) == 'nccl':
+            return self.allgather_cat(tensor.cuda()).cpu()
         out: List[torch.Tensor] = [torch.empty_like(tensor) for _ in range(self.size)]
         dist.all_gather(out, tensor.contiguous(), group=self.group)
         return torch.cat(out, dim=0)
