@@ -40,6 +40,7 @@ def _install_stubs():
     MPI.Comm = _Dummy
     MPI.Group = _Dummy
     MPI.FLOAT = object()
+    MPI.INT64_T = object()
     MPI.SUM = object()
     MPI.LOR = object()
     MPI.COMM_WORLD = None
@@ -205,6 +206,53 @@ def main():
             out[f'load_{ci}_send_{si}_buf'] = buf
         out[f'load_{ci}_n_sends'] = np.asarray([len(comm.sends)])
     out['load_n_cases'] = np.asarray([len(loader_cases)])
+
+    # ---- 5. MatrixSlice single-rank golden --------------------------------
+    # (matrix_slice.py:106-154; P=1 makes the MPI exchanges identities we
+    # can fake, pinning the table construction + localisation fields)
+    class FakeComm1:
+        def Get_rank(self):
+            return 0
+
+        def Get_size(self):
+            return 1
+
+        def allgather(self, x):
+            return [x]
+
+        def alltoall(self, xs):
+            return list(xs)
+
+        def Alltoall(self, send, recv):
+            np.copyto(recv, np.asarray(send))
+
+        def Alltoallv(self, send_spec, recv_spec):
+            recv_spec[0][:send_spec[0].size] = send_spec[0]
+
+        def Barrier(self):
+            pass
+
+    from arrow import matrix_slice as ref_ms
+    ms_cases = []
+    for t in range(4):
+        nn = int(rng.integers(6, 40))
+        dens = float(rng.random() * 0.4)
+        m = sparse.csr_matrix(sparse.random(nn, nn, density=dens,
+                                            random_state=np.random.RandomState(50 + t),
+                                            format='csr'), dtype=np.float32)
+        ms_cases.append(m)
+    for ci, m in enumerate(ms_cases):
+        ms = ref_ms.MatrixSlice.initialize(FakeComm1(), m.copy())
+        out[f'ms_{ci}_dense'] = m.toarray()
+        out[f'ms_{ci}_local'] = ms.A_i_local.toarray()
+        out[f'ms_{ci}_x_index_in'] = np.asarray(ms.x_index_in)
+        out[f'ms_{ci}_rank_in'] = np.asarray(ms.rank_in)
+        out[f'ms_{ci}_x_index_out'] = np.asarray(ms.x_index_out)
+        out[f'ms_{ci}_rank_out'] = np.asarray(ms.rank_out)
+        out[f'ms_{ci}_send_count'] = np.asarray(ms.send_count)
+        out[f'ms_{ci}_recv_count'] = np.asarray(ms.recv_count)
+        out[f'ms_{ci}_bounds'] = np.asarray([ms.start_col, ms.end_col])
+    out['ms_n_cases'] = np.asarray([len(ms_cases)])
 
     path = os.path.join(HERE, 'reference_katsets.npz')
     np.savez_compressed(path, **out)

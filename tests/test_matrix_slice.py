@@ -106,3 +106,23 @@ def test_matrix_slice_gloo_world2():
         assert p.exitcode == 0, f"worker failed (exitcode {p.exitcode})"
     res = q.get(timeout=10)
     assert res == "ok", res
+
+
+def test_single_rank_vs_reference_golden():
+    """Field-by-field pin against the REFERENCE MatrixSlice run at P=1
+    (fixtures from tests/golden/gen_golden.py)."""
+    golden = np.load(os.path.join(os.path.dirname(__file__), 'golden',
+                                  'reference_katsets.npz'))
+    n = int(golden['ms_n_cases'][0])
+    assert n >= 4
+    for ci in range(n):
+        A = sparse.csr_matrix(golden[f'ms_{ci}_dense'].astype(np.float32))
+        ms = MatrixSlice.initialize(None, A.copy())
+        np.testing.assert_array_equal(ms.A_i_local.toarray(),
+                                      golden[f'ms_{ci}_local'])
+        for f in ('x_index_in', 'rank_in', 'x_index_out', 'rank_out',
+                  'send_count', 'recv_count'):
+            np.testing.assert_array_equal(np.asarray(getattr(ms, f)),
+                                          golden[f'ms_{ci}_{f}'])
+        np.testing.assert_array_equal([ms.start_col, ms.end_col],
+                                      golden[f'ms_{ci}_bounds'])
