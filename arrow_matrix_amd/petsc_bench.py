@@ -16,10 +16,11 @@ from .spmm_petsc import SpmmPETSc
 
 
 def load_matrix_slice(some_slice: str, rank: int):
-    """`{name}.part.{x}.slice.{y}.npz` naming (spmm_petsc.py:82-102)."""
+    """`{name}.part.{x}.slice.{y}.npz` naming: replace the second-to-last
+    dot component with this rank (reference spmm_petsc.py:82-102)."""
     parts = some_slice.split('.')
-    base, x = '.'.join(parts[:-4]), parts[-3]
-    return sparse.load_npz(f"{base}.part.{x}.slice.{rank}.npz").tocsr()
+    parts[-2] = str(rank)
+    return sparse.load_npz('.'.join(parts)).tocsr()
 
 
 def benchmark_spmm(matrix_slice_file: Optional[str], k: int, iterations: int,

@@ -111,3 +111,20 @@ def test_petsc_gpu():
     X = rng.rand(500, 32).astype(np.float32)
     Y = eng.spmm(X)
     np.testing.assert_allclose(Y.cpu().numpy(), A @ X, rtol=1e-4, atol=1e-4)
+
+
+def test_benchmark_spmm_slice_files():
+    """Slice-file loading with the reference's naming
+    ({name}.part.{x}.slice.{y}.npz, spmm_petsc.py:82-102)."""
+    import tempfile
+    from arrow_matrix_amd.petsc_bench import benchmark_spmm, load_matrix_slice
+    rng = np.random.RandomState(8)
+    A = sparse.csr_matrix(sparse.random(12, 12, density=0.3, random_state=rng,
+                                        format='csr'), dtype=np.float32)
+    with tempfile.TemporaryDirectory() as td:
+        f = os.path.join(td, 'g.part.1.slice.0.npz')
+        sparse.save_npz(f, A)
+        B = load_matrix_slice(os.path.join(td, 'g.part.1.slice.7.npz'), 0)
+        assert (B != A).nnz == 0
+        Y = benchmark_spmm(f, 3, 1, 'cpu', rng=np.random.default_rng(1))
+        assert Y is not None and np.isfinite(Y.numpy()).all()
