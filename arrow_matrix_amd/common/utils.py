@@ -17,6 +17,33 @@ def str2bool(v: Union[str, bool]) -> bool:
     raise argparse.ArgumentTypeError('Boolean value expected.')
 
 
+def relabel_nodes(g, mapping):
+    """Relabel graph nodes by a mapping (reference utils.py:20-51):
+    g' = I[order] @ g @ I[order]^T."""
+    from scipy import sparse as sp
+    if not isinstance(g, (sp.csr_array, sp.csr_matrix)):
+        raise TypeError("The graph must be a SciPy-compatible CSR array or matrix.")
+    if g.shape[0] != g.shape[1]:
+        raise ValueError("The matrix must be square.")
+    labels = list(range(g.shape[0]))
+    if sorted(mapping.keys()) != labels:
+        raise ValueError("The keys of the mapping must be the rows of the "
+                         "graph's matrix representation.")
+    if sorted(mapping.values()) != labels:
+        raise ValueError("The values of the mapping must be the rows of the "
+                         "graph's matrix representation.")
+    order = [mapping[i] for i in range(g.shape[0])]
+    I = sp.eye(g.shape[0], format='coo', dtype=np.int32)
+    I.row = I.row[order]
+    I = I.tocsr()
+    return I @ g @ I.T
+
+
+def time_to_ms(runtime: float) -> int:
+    """Reference utils.py:54-55."""
+    return int(runtime * 1000)
+
+
 def mpi_print(rank: int, msg: str):
     if rank == 0:
         print(msg, flush=True)
