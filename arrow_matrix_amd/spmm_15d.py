@@ -60,12 +60,15 @@ class Spmm15D:
         NI, NK = A.shape
         self.NJ = X_cols
         self.lNI = int(np.ceil(NI / p_div_c))
-        lNK_panel = int(np.ceil(NK / c))
         self.lNKb = int(np.ceil(NK / p_div_c))
+        # panel width is rounds*lNKb (reference :81 redefines lNK this way)
+        # so sub-block boundaries line up with the bcast roots' X blocks on
+        # ragged sizes
+        lNK_panel = self.rounds * self.lNKb
 
         # my A panel, split into `rounds` column sub-blocks (:124-135)
         r0, r1 = self.x * self.lNI, min(NI, (self.x + 1) * self.lNI)
-        c0, c1 = self.y * lNK_panel, min(NK, (self.y + 1) * lNK_panel)
+        c0, c1 = min(NK, self.y * lNK_panel), min(NK, (self.y + 1) * lNK_panel)
         panel = sparse.csr_matrix(A[r0:r1, c0:c1])
         self.my_rows = (r0, r1)
         self.A_blocks: List = []

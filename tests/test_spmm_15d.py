@@ -49,14 +49,14 @@ def _worker(rank, port, world, c, q):
     try:
         from arrow_matrix_amd.comm import TorchDistComm
         comm = TorchDistComm()
-        n, k = 32, 4
+        n, k = (31 if c == 1 else 32), 4
         A = _rand(n, n, 0.25, 7)
         eng = Spmm15D(comm, A, X_cols=k, c=c, device='cpu')
         rng = np.random.default_rng(9)
         X_full = rng.random((n, k), dtype=np.float32)
         # my replicated X panel: bcast-rank x owns rows [x*lNKb, (x+1)*lNKb)
-        x0 = eng.x * eng.lNKb
-        X_local = X_full[x0:x0 + eng.lNKb]
+        x0 = min(n, eng.x * eng.lNKb)
+        X_local = X_full[x0:min(n, x0 + eng.lNKb)]
         Y = eng.spmm(X_local.copy())
         ref = (A @ X_full)[eng.my_rows[0]:eng.my_rows[1]]
         np.testing.assert_allclose(Y.numpy(), ref, rtol=1e-4, atol=1e-5)
@@ -69,6 +69,16 @@ def _worker(rank, port, world, c, q):
         raise
     finally:
         dist.destroy_process_group()
+
+
+def test_single_process_ragged():
+    # n not divisible by the grid: panel/sub-block clipping paths
+    A = _rand(25, 25, 0.3, 5)
+    eng = Spmm15D(None, A, X_cols=3, c=1, device='cpu')
+    rng = np.random.default_rng(1)
+    X = rng.random((25, 3), dtype=np.float32)
+    Y = eng.spmm(X)
+    np.testing.assert_allclose(Y.numpy(), A @ X, rtol=1e-5, atol=1e-6)
 
 
 @pytest.mark.parametrize("world,c", [(2, 1), (4, 2)])
