@@ -119,3 +119,30 @@ def test_number_of_blocks_vs_golden(golden):
 def test_number_of_blocks_all_zero_raises():
     with pytest.raises(ValueError):
         tables.number_of_blocks(np.zeros(10, dtype=np.int64), 2)
+
+
+def test_golden_fixtures_are_fresh(tmp_path, monkeypatch):
+    """When the reference tree is available (build container), regenerating
+    the golden fixtures must reproduce the committed ones — guards against
+    fixture drift after generator or synth changes. Skipped on boxes
+    without /root/reference."""
+    import importlib
+    import subprocess
+    import sys
+    ref = os.environ.get('ARROW_REFERENCE_PATH', '/root/reference')
+    if not os.path.isdir(ref):
+        pytest.skip("reference tree not available")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    gen = os.path.join(repo, 'tests', 'golden', 'gen_golden.py')
+    out_dir = tmp_path / 'golden'
+    out_dir.mkdir()
+    import shutil
+    shutil.copy(gen, out_dir / 'gen_golden.py')
+    r = subprocess.run([sys.executable, str(out_dir / 'gen_golden.py')],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    fresh = np.load(out_dir / 'reference_katsets.npz')
+    committed = np.load(GOLDEN)
+    assert sorted(fresh.files) == sorted(committed.files)
+    for key in committed.files:
+        np.testing.assert_array_equal(fresh[key], committed[key], err_msg=key)
