@@ -254,7 +254,8 @@ def main():
         out[f'ms_{ci}_bounds'] = np.asarray([ms.start_col, ms.end_col])
     out['ms_n_cases'] = np.asarray([len(ms_cases)])
 
-    path = os.path.join(HERE, 'reference_katsets.npz')
+    path = os.environ.get('ARROW_GOLDEN_OUT',
+                          os.path.join(HERE, 'reference_katsets.npz'))
     np.savez_compressed(path, **out)
     print(f'wrote {path} ({os.path.getsize(path)/1024:.1f} KiB, {len(out)} arrays)')
 

@@ -126,7 +126,6 @@ def test_golden_fixtures_are_fresh(tmp_path, monkeypatch):
     the golden fixtures must reproduce the committed ones — guards against
     fixture drift after generator or synth changes. Skipped on boxes
     without /root/reference."""
-    import importlib
     import subprocess
     import sys
     ref = os.environ.get('ARROW_REFERENCE_PATH', '/root/reference')
@@ -134,14 +133,12 @@ def test_golden_fixtures_are_fresh(tmp_path, monkeypatch):
         pytest.skip("reference tree not available")
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     gen = os.path.join(repo, 'tests', 'golden', 'gen_golden.py')
-    out_dir = tmp_path / 'golden'
-    out_dir.mkdir()
-    import shutil
-    shutil.copy(gen, out_dir / 'gen_golden.py')
-    r = subprocess.run([sys.executable, str(out_dir / 'gen_golden.py')],
-                       capture_output=True, text=True, timeout=300)
+    out = tmp_path / 'fresh.npz'
+    env = dict(os.environ, ARROW_GOLDEN_OUT=str(out))
+    r = subprocess.run([sys.executable, gen], capture_output=True, text=True,
+                       timeout=300, env=env)
     assert r.returncode == 0, r.stderr[-1500:]
-    fresh = np.load(out_dir / 'reference_katsets.npz')
+    fresh = np.load(out)
     committed = np.load(GOLDEN)
     assert sorted(fresh.files) == sorted(committed.files)
     for key in committed.files:
