@@ -180,7 +180,7 @@ class ArrowSlimMPI(ArrowMatrix):
         w = self.width
         nw = self.n_owned
         self._A_row0 = None
-        self._A_rest = None
+        self._A_rest = []
         self._A_all = None
         self._rest_row_offset = 0
         if nw == 0:
@@ -285,14 +285,49 @@ class ArrowSlimMPI(ArrowMatrix):
                                               x_rows=nw * w)
         elif n_rest:
             rest_rows = nw * w - self._rest_row_offset
-            self._A_rest = self._merged_handle(rest_rows, nw * w, rows_cat,
+            x_rows_total = (n_rest * w + w if not split_col else n_rest * w)
+            # Auto-split into row-range chunks when the merged nonzero count
+            # would overflow the per-structure int32 limit (scale headroom
+            # beyond cfg4); ARROW_REST_CHUNK_NNZ forces small chunks in tests.
+            cap = int(_os.environ.get('ARROW_REST_CHUNK_NNZ', 1 << 30))
+            self._A_rest = self._merged_chunks(rest_rows, nw * w, rows_cat,
                                                cols_cat, data_cat,
-                                               x_rows=(n_rest * w + w
-                                                       if not split_col else n_rest * w))
+                                               x_rows_total, cap)
             if split_col and col_rows:
                 self._A_col = self._merged_col_sorted(rest_rows, nw * w,
                                                       col_rows, col_cols,
                                                       col_data)
+
+    def _merged_chunks(self, n_rows, n_cols, rows_cat, cols_cat, data_cat,
+                       x_rows_total, cap):
+        """One merged structure, or several row-range chunks when the
+        nonzero count would overflow the per-structure int32 limit."""
+        rows = np.concatenate(rows_cat)
+        total_nnz = rows.size
+        if total_nnz <= cap:
+            h = self._merged_handle(n_rows, n_cols, [rows], cols_cat,
+                                    data_cat, x_rows=x_rows_total)
+            return [(h, 0, n_rows)]
+        cols = np.concatenate(cols_cat)
+        data = np.concatenate(data_cat)
+        per_row = np.bincount(rows, minlength=n_rows)
+        cum = np.cumsum(per_row)
+        n_chunks = int(np.ceil(total_nnz / cap))
+        out = []
+        lo = 0
+        for q in range(n_chunks):
+            target = total_nnz * (q + 1) // n_chunks
+            hi = int(np.searchsorted(cum, target)) + 1 if q < n_chunks - 1 else n_rows
+            hi = max(min(hi, n_rows), lo + 1)
+            m = (rows >= lo) & (rows < hi)
+            h = self._merged_handle(hi - lo, n_cols, [rows[m] - lo],
+                                    [cols[m]], [data[m]],
+                                    x_rows=max(1, x_rows_total // n_chunks))
+            out.append((h, lo, hi))
+            lo = hi
+            if lo >= n_rows:
+                break
+        return out
 
     def _merged_col_sorted(self, n_rows, n_cols, rows_cat, cols_cat, data_cat):
         """Build the X_0-entry structure with ROW BLOCKS ordered by their
@@ -533,12 +568,13 @@ class ArrowSlimMPI(ArrowMatrix):
         # arrow_slim_mpi.py:121-144)
         if bcast_work is not None:
             bcast_work.wait()  # rest reads X_0
-        if self._A_rest is not None:
-            h = self._A_rest
-            C_sub = self.C_i[self._rest_row_offset:
-                             self.n_owned * w]
-            self._timed(lambda: be.spmm_dual(h, self.X_i, self.X_0, C_sub, 0),
-                        h.nnz, C_sub.shape[0], h.x_rows)
+        if self._A_rest:
+            for h, lo, hi in self._A_rest:
+                C_sub = self.C_i[self._rest_row_offset + lo:
+                                 self._rest_row_offset + hi]
+                self._timed(lambda h=h, C_sub=C_sub:
+                            be.spmm_dual(h, self.X_i, self.X_0, C_sub, 0),
+                            h.nnz, C_sub.shape[0], h.x_rows)
         if self._A_col is not None:
             h = self._A_col
             C_full = self.C_i[self._rest_row_offset:self.n_owned * w]

@@ -94,8 +94,10 @@ def test_merged_structures_match_per_block(nb, w, first, last):
     np.testing.assert_allclose(C0, C0_ref, rtol=1e-5, atol=1e-6)
 
     # rest merged: C_r = A_rr @ X_r + A_r0 @ X_0
-    if eng._A_rest is not None:
-        Crest = _apply_merged(eng._A_rest, X_stripe, X_0)
+    if eng._A_rest:
+        Crest = np.zeros((eng._A_rest[-1][2], k), np.float32)
+        for h, lo, hi in eng._A_rest:
+            Crest[lo:hi] = _apply_merged(h, X_stripe, X_0)
         off = eng._rest_row_offset
         for j, r in enumerate(range(first, last)):
             if r == 0:
@@ -135,7 +137,9 @@ def test_split_col_structure(monkeypatch):
     k = 3
     X_stripe = (2 * rng.random(((last - first) * w, k)) - 1).astype(np.float32)
     X_0 = (2 * rng.random((w, k)) - 1).astype(np.float32)
-    C = _apply_merged(eng._A_rest, X_stripe, X_0)
+    C = np.zeros((eng._A_rest[-1][2], k), np.float32)
+    for rh, lo, hi in eng._A_rest:
+        C[lo:hi] = _apply_merged(rh, X_stripe, X_0)
     # add the col structure with its explicit row ids
     for r in range(h.shape[0]):
         out = int(h.row_ids[r])
