@@ -185,3 +185,18 @@ def test_allreduce_x0_multi_part_invalidation():
             np.testing.assert_allclose(C, golden, rtol=5e-4, atol=5e-4)
             golden_X = compute_spmm(decomp, golden_X)
             arrow.B.set_features(arrow.B.result_tile())
+
+
+@pytest.mark.gpu
+def test_engine_gpu_rest_chunked(monkeypatch):
+    """Force multi-chunk rest structures (int32-overflow headroom path)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    monkeypatch.setenv('ARROW_REST_CHUNK_NNZ', '500')
+    from tests.test_engine_cpu import _run_engine
+    from arrow_matrix_amd import synth
+    decomp = synth.synth_arrow_decomposition(64, [4], avg_deg=8, seed=29)
+    results, goldens = _run_engine(decomp, 64, [4], 16, iters=2,
+                                   device='gpu', seed=29)
+    for C, G in zip(results, goldens):
+        np.testing.assert_allclose(C, G, rtol=2e-4, atol=2e-4)
