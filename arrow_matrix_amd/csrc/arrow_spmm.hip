@@ -70,7 +70,19 @@ struct CsrBlock {
   // XCD-contiguous item remap for this structure (uniform banded rows only:
   // on hub-heavy structures it serialises the heavy head onto one XCD)
   int xcd_remap = 0;
+  // per-XCD queue scheduler: -1 follow ARROW_QUEUE env, 0 off, 1 on
+  int queue_mode = -1;
+  int64_t *qseg = nullptr;  // device, 9 nnz-balanced item-segment bounds
+  int32_t *qctr = nullptr;  // device, 8 chunk counters padded 32 ints apart
 };
+
+int env_queue_default() {
+  static const int v = [] {
+    const char *e = getenv("ARROW_QUEUE");
+    return (e && e[0] == '1') ? 1 : 0;
+  }();
+  return v;
+}
 
 std::unordered_map<int64_t, CsrBlock> g_blocks;
 int64_t g_next_handle = 1;
@@ -78,60 +90,19 @@ int64_t g_next_handle = 1;
 // ---------------------------------------------------------------------------
 // SpMM kernel.  GROUP lanes x VEC floats cover min(k, GROUP*VEC) columns;
 // wider k is handled by a column-offset loop over launches (col_off).
+// The per-item body is shared by the two schedulers below (grid-stride and
+// per-XCD queue) — spmm_process_item computes one work item end to end.
 // ---------------------------------------------------------------------------
 
 template <int VEC, int GROUP, int BETA, bool GUARD>
-__global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
-    const int2 *__restrict__ pairs,
-    const int32_t *__restrict__ item_row, const int32_t *__restrict__ item_begin,
-    const int32_t *__restrict__ item_end, int64_t n_items,
+__device__ __forceinline__ void spmm_process_item(
+    const int2 *__restrict__ pairs, int32_t row_raw, int32_t b, int32_t e,
     const float *__restrict__ X0, const float *__restrict__ X1,
-    float *__restrict__ C, int64_t k, int64_t col_off, int xcd_remap,
-    int nt_mode) {
-  constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
-  const int lane_in_group = threadIdx.x % GROUP;
-  const int group_in_block = threadIdx.x / GROUP;
-  const int64_t col0 = col_off + (int64_t)lane_in_group * VEC;
-  const bool active = !GUARD || (col0 < k);
-
-  // XCD-aware workgroup remap (performance only): the dispatcher places
-  // block b on XCD b%8, so remap block ids to give each XCD a CONTIGUOUS
-  // range of work items — consecutive rows of a banded block then share the
-  // XCD's private L2 window instead of interleaving across all 8 L2s.
-  // Bijective form (cdna_hip_programming.md §XCD swizzle).
-  int wg = blockIdx.x;
-  if (xcd_remap) {
-    const int nwg = gridDim.x;
-    const int q = nwg / 8, rm = nwg % 8;
-    const int xcd = blockIdx.x % 8, pos = blockIdx.x / 8;
-    wg = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + pos;
-  }
-
-  int64_t item = (int64_t)wg * GROUPS_PER_BLOCK + group_in_block;
-  const int64_t stride = (int64_t)gridDim.x * GROUPS_PER_BLOCK;
-
-  // software prefetch of the NEXT item's metadata: without it every ~8-nnz
-  // item pays a 3-deep dependent load chain (meta -> pairs -> X row) that
-  // dominates short power-law rows
-  int32_t next_row = 0, next_b = 0, next_e = 0;
-  if (item < n_items) {
-    next_row = __builtin_nontemporal_load(&item_row[item]);
-    next_b = __builtin_nontemporal_load(&item_begin[item]);
-    next_e = __builtin_nontemporal_load(&item_end[item]);
-  }
-
-  for (; item < n_items; item += stride) {
-    const int32_t row_raw = next_row;
+    float *__restrict__ C, int64_t k, int64_t col0, bool active,
+    int lane_in_group, int nt_mode) {
+  {
     const int32_t row = row_raw & 0x7fffffff;
     const bool is_split = row_raw < 0;
-    const int32_t b = next_b;
-    const int32_t e = next_e;
-    const int64_t nxt = item + stride;
-    if (nxt < n_items) {
-      next_row = __builtin_nontemporal_load(&item_row[nxt]);
-      next_b = __builtin_nontemporal_load(&item_begin[nxt]);
-      next_e = __builtin_nontemporal_load(&item_end[nxt]);
-    }
 
     float acc[VEC];
 #pragma unroll
@@ -228,11 +199,20 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
         consume4(mine, u + 4);
       }
     };
+    // nontemporal A-stream load (each pair is read once by one wave): the
+    // builtin wants a native vector type, HIP's int2 is a class
+    typedef int nt_int2 __attribute__((ext_vector_type(2)));
+    auto load_pair = [&](int32_t idx) -> int2 {
+      if (nt_mode) {
+        const nt_int2 raw = __builtin_nontemporal_load(
+            reinterpret_cast<const nt_int2 *>(pairs + idx));
+        return int2{raw.x, raw.y};
+      }
+      return pairs[idx];
+    };
     int32_t base = b;
     for (; base + GROUP <= e; base += GROUP) {  // full chunks
-      const int2 mine = nt_mode
-          ? __builtin_nontemporal_load(&pairs[base + lane_in_group])
-          : pairs[base + lane_in_group];
+      const int2 mine = load_pair(base + lane_in_group);
       if constexpr (GROUP >= 8) {
 #pragma unroll
         for (int u = 0; u < GROUP; u += 8) consume8(mine, u);
@@ -272,16 +252,148 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
         }
       } else {
         if constexpr (VEC == 4) {
-          float4 outv{acc[0], acc[1], acc[2], acc[3]};
           if (nt_mode) {
-            __builtin_nontemporal_store(outv, reinterpret_cast<float4 *>(cr));
+            typedef float nt_float4 __attribute__((ext_vector_type(4)));
+            const nt_float4 outv = {acc[0], acc[1], acc[2], acc[3]};
+            __builtin_nontemporal_store(outv,
+                                        reinterpret_cast<nt_float4 *>(cr));
           } else {
-            *reinterpret_cast<float4 *>(cr) = outv;
+            *reinterpret_cast<float4 *>(cr) = float4{acc[0], acc[1], acc[2],
+                                                     acc[3]};
           }
         } else {
 #pragma unroll
           for (int j = 0; j < VEC; ++j) cr[j] = acc[j];
         }
+      }
+    }
+  }
+}
+
+template <int VEC, int GROUP, int BETA, bool GUARD>
+__global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel(
+    const int2 *__restrict__ pairs,
+    const int32_t *__restrict__ item_row, const int32_t *__restrict__ item_begin,
+    const int32_t *__restrict__ item_end, int64_t n_items,
+    const float *__restrict__ X0, const float *__restrict__ X1,
+    float *__restrict__ C, int64_t k, int64_t col_off, int xcd_remap,
+    int nt_mode) {
+  constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
+  const int lane_in_group = threadIdx.x % GROUP;
+  const int group_in_block = threadIdx.x / GROUP;
+  const int64_t col0 = col_off + (int64_t)lane_in_group * VEC;
+  const bool active = !GUARD || (col0 < k);
+
+  // XCD-aware workgroup remap (performance only): the dispatcher places
+  // block b on XCD b%8, so remap block ids to give each XCD a CONTIGUOUS
+  // range of work items — consecutive rows of a banded block then share the
+  // XCD's private L2 window instead of interleaving across all 8 L2s.
+  // Bijective form (cdna_hip_programming.md §XCD swizzle).
+  int wg = blockIdx.x;
+  if (xcd_remap) {
+    const int nwg = gridDim.x;
+    const int q = nwg / 8, rm = nwg % 8;
+    const int xcd = blockIdx.x % 8, pos = blockIdx.x / 8;
+    wg = (xcd < rm ? xcd * (q + 1) : rm * (q + 1) + (xcd - rm) * q) + pos;
+  }
+
+  int64_t item = (int64_t)wg * GROUPS_PER_BLOCK + group_in_block;
+  const int64_t stride = (int64_t)gridDim.x * GROUPS_PER_BLOCK;
+
+  // software prefetch of the NEXT item's metadata: without it every ~8-nnz
+  // item pays a 3-deep dependent load chain (meta -> pairs -> X row) that
+  // dominates short power-law rows
+  int32_t next_row = 0, next_b = 0, next_e = 0;
+  if (item < n_items) {
+    next_row = __builtin_nontemporal_load(&item_row[item]);
+    next_b = __builtin_nontemporal_load(&item_begin[item]);
+    next_e = __builtin_nontemporal_load(&item_end[item]);
+  }
+
+  for (; item < n_items; item += stride) {
+    const int32_t row_raw = next_row;
+    const int32_t b = next_b;
+    const int32_t e = next_e;
+    const int64_t nxt = item + stride;
+    if (nxt < n_items) {
+      next_row = __builtin_nontemporal_load(&item_row[nxt]);
+      next_b = __builtin_nontemporal_load(&item_begin[nxt]);
+      next_e = __builtin_nontemporal_load(&item_end[nxt]);
+    }
+    spmm_process_item<VEC, GROUP, BETA, GUARD>(
+        pairs, row_raw, b, e, X0, X1, C, k, col0, active, lane_in_group,
+        nt_mode);
+  }
+}
+
+// Per-XCD queue scheduler. The static grid-stride schedule above interleaves
+// consecutive work items across all 8 XCDs and lets workgroups DRIFT apart
+// over millions of items, so the ~10 consumers of each 512-B X row (rows
+// within ±band) hit different L2s at different times — measured 3.2x X
+// re-fetch, 16 % L2 hit (profiles/r01_pmc_sq_tcc_20M.txt). Here items are
+// split into 8 CONTIGUOUS nnz-balanced row segments; each workgroup drains
+// the segment of the XCD it actually runs on (HW_REG_XCC_ID, speed-only)
+// via an atomic chunk counter, so every XCD walks one tight row window in
+// order: consumers of an X row share one private L2, and the in-flight
+// window stays a few MB. Workgroups that exhaust their queue steal from the
+// next (placement-independent correctness; the register read is only a
+// locality hint — cdna_hip_programming.md §Guideline 16).
+__device__ __forceinline__ unsigned arrow_xcc_id() {
+  // s_getreg_b32 HW_REG_XCC_ID: reg 20, offset 0, size 4 (gfx950)
+  return __builtin_amdgcn_s_getreg((3 << 11) | (0 << 6) | 20);
+}
+
+template <int VEC, int GROUP, int BETA, bool GUARD>
+__global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel_q(
+    const int2 *__restrict__ pairs,
+    const int32_t *__restrict__ item_row, const int32_t *__restrict__ item_begin,
+    const int32_t *__restrict__ item_end,
+    const int64_t *__restrict__ qseg,  // 9 segment bounds (items)
+    int32_t *__restrict__ qctr,       // 8 counters, padded 32 ints apart
+    int chunk_items,
+    const float *__restrict__ X0, const float *__restrict__ X1,
+    float *__restrict__ C, int64_t k, int64_t col_off, int nt_mode) {
+  constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
+  const int lane_in_group = threadIdx.x % GROUP;
+  const int group_in_block = threadIdx.x / GROUP;
+  const int64_t col0 = col_off + (int64_t)lane_in_group * VEC;
+  const bool active = !GUARD || (col0 < k);
+  __shared__ int64_t s_base;
+
+  const unsigned q0 = arrow_xcc_id() & 7;
+  for (unsigned qi = 0; qi < 8; ++qi) {
+    const unsigned q = (q0 + qi) & 7;
+    const int64_t lo = qseg[q], hi = qseg[q + 1];
+    if (lo >= hi) continue;
+    for (;;) {
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        s_base = lo + (int64_t)atomicAdd(&qctr[q * 32], chunk_items);
+      }
+      __syncthreads();
+      const int64_t base = s_base;
+      if (base >= hi) break;
+      const int64_t end = base + chunk_items < hi ? base + chunk_items : hi;
+      int64_t item = base + group_in_block;
+      int32_t next_row = 0, next_b = 0, next_e = 0;
+      if (item < end) {
+        next_row = __builtin_nontemporal_load(&item_row[item]);
+        next_b = __builtin_nontemporal_load(&item_begin[item]);
+        next_e = __builtin_nontemporal_load(&item_end[item]);
+      }
+      for (; item < end; item += GROUPS_PER_BLOCK) {
+        const int32_t row_raw = next_row;
+        const int32_t b = next_b;
+        const int32_t e = next_e;
+        const int64_t nxt = item + GROUPS_PER_BLOCK;
+        if (nxt < end) {
+          next_row = __builtin_nontemporal_load(&item_row[nxt]);
+          next_b = __builtin_nontemporal_load(&item_begin[nxt]);
+          next_e = __builtin_nontemporal_load(&item_end[nxt]);
+        }
+        spmm_process_item<VEC, GROUP, BETA, GUARD>(
+            pairs, row_raw, b, e, X0, X1, C, k, col0, active, lane_in_group,
+            nt_mode);
       }
     }
   }
@@ -441,8 +553,24 @@ template <int VEC, int GROUP>
 int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
                    float *C, int64_t k, int beta, hipStream_t stream) {
   constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
+  // A/B: non-temporal pairs load + C store (measured ~0 at default shapes)
+  static const int nt_mode = [] {
+    const char *e = getenv("ARROW_SPMM_NT");
+    return (e && e[0] == '1') ? 1 : 0;
+  }();
+  const bool useq = (blk.queue_mode >= 0 ? blk.queue_mode
+                                         : env_queue_default()) && blk.qseg;
+  // Queue mode: size the grid to residency (8 blocks/CU fit at this
+  // occupancy — 4 waves/WG, 8 waves/SIMD), not to the item count; chunk =
+  // 2 rounds per grab keeps each XCD's in-flight row window a few MB.
+  static const int q_chunk_mult = [] {
+    const char *e = getenv("ARROW_Q_CHUNK");
+    return e ? std::max(1, atoi(e)) : 2;
+  }();
+  const int chunk_items = GROUPS_PER_BLOCK * q_chunk_mult;
   int blocks = (int)std::min<int64_t>(
-      (blk.n_items + GROUPS_PER_BLOCK - 1) / GROUPS_PER_BLOCK, 8192);
+      (blk.n_items + GROUPS_PER_BLOCK - 1) / GROUPS_PER_BLOCK,
+      useq ? 2048 : 8192);
   if (blocks < 1) blocks = 1;
   const int64_t span = (int64_t)GROUP * VEC;
   for (int64_t col_off = 0; col_off < k; col_off += span) {
@@ -453,7 +581,22 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
                          blk.item_end, blk.n_items, X0, X1, C, k, col_off,
                          blk.xcd_remap, nt_mode);
     };
-    if (beta == 0) {
+    auto runq = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(blocks), dim3(BLOCK_THREADS), 0, stream,
+                         blk.pairs, blk.item_row, blk.item_begin,
+                         blk.item_end, blk.qseg, blk.qctr, chunk_items,
+                         X0, X1, C, k, col_off, nt_mode);
+    };
+    if (useq) {
+      HIP_CHECK(hipMemsetAsync(blk.qctr, 0, 8 * 32 * sizeof(int32_t), stream));
+      if (beta == 0) {
+        if (guard) runq(spmm_kernel_q<VEC, GROUP, 0, true>);
+        else       runq(spmm_kernel_q<VEC, GROUP, 0, false>);
+      } else {
+        if (guard) runq(spmm_kernel_q<VEC, GROUP, 1, true>);
+        else       runq(spmm_kernel_q<VEC, GROUP, 1, false>);
+      }
+    } else if (beta == 0) {
       if (guard) run(spmm_kernel<VEC, GROUP, 0, true>);
       else       run(spmm_kernel<VEC, GROUP, 0, false>);
     } else {
@@ -548,6 +691,25 @@ static int64_t csr_create_impl(int64_t rows, int64_t cols, int64_t nnz,
   blk.n_items = (int64_t)item_row.size();
   blk.n_split_rows = (int64_t)split_rows.size();
 
+  // nnz-balanced contiguous item segments for the per-XCD queue scheduler
+  // (items are in row order, so a segment is a contiguous row window)
+  int64_t qseg_h[9];
+  {
+    // weight = nnz + 1 per item: balances by data volume while keeping
+    // empty/short rows (fixed per-item overhead) spread across queues
+    const int64_t n_it = blk.n_items;
+    const int64_t total_w = nnz + n_it;
+    int64_t cum = 0, target_q = 1;
+    qseg_h[0] = 0;
+    for (int64_t i = 0; i < n_it && target_q < 8; ++i) {
+      cum += (item_end[(size_t)i] - item_begin[(size_t)i]) + 1;
+      while (target_q < 8 && cum * 8 >= total_w * target_q) {
+        qseg_h[target_q++] = i + 1;
+      }
+    }
+    while (target_q <= 8) qseg_h[target_q++] = n_it;
+  }
+
   // pack (col, val) into 8-byte pairs for the staged kernel loads
   std::vector<int2> pairs((size_t)nnz);
   for (int64_t t = 0; t < nnz; ++t) {
@@ -566,9 +728,11 @@ static int64_t csr_create_impl(int64_t rows, int64_t cols, int64_t nnz,
       upload((void **)&blk.item_row, item_row.data(), item_row.size() * 4) ||
       upload((void **)&blk.item_begin, item_begin.data(), item_begin.size() * 4) ||
       upload((void **)&blk.item_end, item_end.data(), item_end.size() * 4) ||
-      upload((void **)&blk.split_rows, split_rows.data(), split_rows.size() * 4)) {
+      upload((void **)&blk.split_rows, split_rows.data(), split_rows.size() * 4) ||
+      upload((void **)&blk.qseg, qseg_h, sizeof(qseg_h))) {
     return -1;
   }
+  HIP_CHECK(hipMalloc((void **)&blk.qctr, 8 * 32 * sizeof(int32_t)));
   const int64_t h = g_next_handle++;
   g_blocks.emplace(h, blk);
   return h;
@@ -595,7 +759,7 @@ int arrow_csr_destroy(int64_t handle) {
   CsrBlock &b = it->second;
   for (void *p : {(void *)b.pairs, (void *)b.item_row,
                   (void *)b.item_begin, (void *)b.item_end,
-                  (void *)b.split_rows}) {
+                  (void *)b.split_rows, (void *)b.qseg, (void *)b.qctr}) {
     if (p) (void)hipFree(p);
   }
   g_blocks.erase(it);
@@ -615,6 +779,16 @@ int arrow_csr_set_xcd_remap(int64_t handle, int enable) {
     return -1;
   }
   it->second.xcd_remap = enable ? 1 : 0;
+  return 0;
+}
+
+int arrow_csr_set_queue(int64_t handle, int mode) {
+  auto it = g_blocks.find(handle);
+  if (it == g_blocks.end()) {
+    set_error("arrow_csr_set_queue: bad handle");
+    return -1;
+  }
+  it->second.queue_mode = mode < 0 ? -1 : (mode ? 1 : 0);
   return 0;
 }
 

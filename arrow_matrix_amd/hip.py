@@ -51,6 +51,8 @@ def _load():
     lib.arrow_csr_nnz.restype = ctypes.c_int64
     lib.arrow_csr_set_xcd_remap.argtypes = [ctypes.c_int64, ctypes.c_int]
     lib.arrow_csr_set_xcd_remap.restype = ctypes.c_int
+    lib.arrow_csr_set_queue.argtypes = [ctypes.c_int64, ctypes.c_int]
+    lib.arrow_csr_set_queue.restype = ctypes.c_int
     lib.arrow_spmm.argtypes = [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
                                ctypes.c_int64, ctypes.c_int, ctypes.c_void_p]
     lib.arrow_spmm.restype = ctypes.c_int
@@ -130,6 +132,11 @@ class CsrBlockGPU:
     def set_xcd_remap(self, enable: bool):
         _check(_load().arrow_csr_set_xcd_remap(self._handle, 1 if enable else 0),
                "arrow_csr_set_xcd_remap")
+
+    def set_queue(self, mode: int):
+        """Per-XCD queue scheduler: 1 on, 0 off, -1 follow ARROW_QUEUE env."""
+        _check(_load().arrow_csr_set_queue(self._handle, int(mode)),
+               "arrow_csr_set_queue")
 
     def spmm_dual(self, X0_ptr: int, X1_ptr: int, C_ptr: int, k: int,
                   beta: int, stream: int = 0):

@@ -63,6 +63,11 @@ int64_t arrow_csr_nnz(int64_t handle);
 /* Enable the XCD-contiguous work remap for this structure (a performance
  * hint for uniform banded rows; off by default). */
 int arrow_csr_set_xcd_remap(int64_t handle, int enable);
+/* Per-XCD queue scheduler for this structure: workgroups drain contiguous
+ * nnz-balanced row segments via per-XCD atomic chunk counters (keeps the
+ * consumers of each X row in ONE XCD's L2). mode: 1 on, 0 off,
+ * -1 follow the ARROW_QUEUE env default. */
+int arrow_csr_set_queue(int64_t handle, int mode);
 
 /* C (+)= A @ X.  X: (cols, k) fp32 row-major device;  C: (rows, k).
  * beta = 0: C = A@X (rows not touched by A are zeroed);  beta = 1: C += A@X. */
