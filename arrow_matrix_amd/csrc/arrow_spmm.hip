@@ -77,9 +77,11 @@ struct CsrBlock {
 };
 
 int env_queue_default() {
+  // Default ON: measured 2612 -> 4026 GF/s at cfg4 (profiles/
+  // r01_bench_queue_on.json). ARROW_QUEUE=0 restores grid-stride.
   static const int v = [] {
     const char *e = getenv("ARROW_QUEUE");
-    return (e && e[0] == '1') ? 1 : 0;
+    return (e && e[0] == '0') ? 0 : 1;
   }();
   return v;
 }
