@@ -347,7 +347,8 @@ def main():
             "unit": "GB/s",
             "frac": round(achieved / peak, 4),
             "traffic": traffic,
-            "kernel": "spmm_kernel",
+            "kernel": ("spmm_kernel" if os.environ.get('ARROW_QUEUE') == '0'
+                       else "spmm_kernel_q"),
             "launches": len(events),
             "avg_launch_ms": round(total_ms / len(events), 4),
         }
