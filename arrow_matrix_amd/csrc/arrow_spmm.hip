@@ -77,11 +77,15 @@ struct CsrBlock {
 };
 
 int env_queue_default() {
-  // Default ON: measured 2612 -> 4026 GF/s at cfg4 (profiles/
-  // r01_bench_queue_on.json). ARROW_QUEUE=0 restores grid-stride.
+  // -1 unset (per-launch policy: queue when GROUP >= 16 — measured
+  // +54 % at k=128/cfg4, +18 % at k=64, but -19 % at k=16 where the
+  // per-chunk barrier dominates 4-lane groups; profiles/
+  // r01_queue_chunk_sweep.txt, r01_ksweep_queue_20M.txt);
+  // ARROW_QUEUE=0 forces grid-stride, ARROW_QUEUE=1 forces queues.
   static const int v = [] {
     const char *e = getenv("ARROW_QUEUE");
-    return (e && e[0] == '0') ? 0 : 1;
+    if (!e || !e[0]) return -1;
+    return (e[0] == '0') ? 0 : 1;
   }();
   return v;
 }
@@ -560,8 +564,8 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
     const char *e = getenv("ARROW_SPMM_NT");
     return (e && e[0] == '1') ? 1 : 0;
   }();
-  const bool useq = (blk.queue_mode >= 0 ? blk.queue_mode
-                                         : env_queue_default()) && blk.qseg;
+  const int qd = blk.queue_mode >= 0 ? blk.queue_mode : env_queue_default();
+  const bool useq = blk.qseg && (qd >= 0 ? qd : (GROUP >= 16));
   // Queue mode: size the grid to residency (8 blocks/CU fit at this
   // occupancy — 4 waves/WG, 8 waves/SIMD), not to the item count; chunk =
   // 2 rounds per grab keeps each XCD's in-flight row window a few MB.
@@ -569,10 +573,14 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
     const char *e = getenv("ARROW_Q_CHUNK");
     return e ? std::max(1, atoi(e)) : 2;
   }();
+  static const int q_blocks = [] {
+    const char *e = getenv("ARROW_Q_BLOCKS");
+    return e ? std::max(8, atoi(e)) : 2048;
+  }();
   const int chunk_items = GROUPS_PER_BLOCK * q_chunk_mult;
   int blocks = (int)std::min<int64_t>(
       (blk.n_items + GROUPS_PER_BLOCK - 1) / GROUPS_PER_BLOCK,
-      useq ? 2048 : 8192);
+      useq ? q_blocks : 8192);
   if (blocks < 1) blocks = 1;
   const int64_t span = (int64_t)GROUP * VEC;
   for (int64_t col_off = 0; col_off < k; col_off += span) {

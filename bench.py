@@ -145,6 +145,22 @@ def build_blocks_for_rank(rank, world, w, nb, parts, gen_device, band):
     return grids, first, last
 
 
+def _kernel_name(k):
+    """Mirror the launcher's scheduler policy (csrc/arrow_spmm.hip): queue
+    scheduler when GROUP >= 16 unless ARROW_QUEUE forces it."""
+    env = os.environ.get('ARROW_QUEUE', '')
+    if env == '0':
+        return "spmm_kernel"
+    if env.startswith('1'):
+        return "spmm_kernel_q"
+    vec = 4 if k % 4 == 0 else (2 if k % 2 == 0 else 1)
+    lanes = (k + vec - 1) // vec
+    group = 1
+    while group < lanes and group < 64:
+        group <<= 1
+    return "spmm_kernel_q" if group >= 16 else "spmm_kernel"
+
+
 def cpu_baseline_sample(w, band, k, threads):
     """Time the reference CPU arithmetic (scipy CSR @ dense,
     arrow_slim_mpi.py:109-144) on one diagonal block of the same workload."""
@@ -347,8 +363,7 @@ def main():
             "unit": "GB/s",
             "frac": round(achieved / peak, 4),
             "traffic": traffic,
-            "kernel": ("spmm_kernel" if os.environ.get('ARROW_QUEUE') == '0'
-                       else "spmm_kernel_q"),
+            "kernel": _kernel_name(k),
             "launches": len(events),
             "avg_launch_ms": round(total_ms / len(events), 4),
         }
