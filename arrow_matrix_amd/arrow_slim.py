@@ -176,6 +176,14 @@ class ArrowSlimMPI(ArrowMatrix):
             n_chunks = max(1, min(int(env_c), w))
         return n_chunks
 
+    @property
+    def _row0_col_items(self) -> bool:
+        """Order the row-0 (hub) structures' work items by column so a
+        per-XCD queue segment covers a column window of X (A/B knob;
+        see arrow_csr_create_opts in include/arrow_spmm.h)."""
+        import os as _os
+        return _os.environ.get('ARROW_ROW0_COLSORT', '0') == '1'
+
     def _build_merged_gpu(self) -> None:
         w = self.width
         nw = self.n_owned
@@ -222,7 +230,8 @@ class ArrowSlimMPI(ArrowMatrix):
                 m = (rows >= lo) & (rows < hi)
                 h = self._merged_handle(hi - lo, nw * w,
                                         [rows[m] - lo], [cols[m]], [data[m]],
-                                        x_rows=nw * w // n_chunks)
+                                        x_rows=nw * w // n_chunks,
+                                        col_items=self._row0_col_items)
                 self._A_row0.append((h, lo, hi))
         # --- rest merge: C[r] = A_rr @ X_r + A_r0 @ X_0
         #     (+ interior banded off-diagonals A_{r,r±1} @ X_{r±1} when the
@@ -358,7 +367,7 @@ class ArrowSlimMPI(ArrowMatrix):
         return handle
 
     def _merged_handle(self, n_rows, n_cols, rows_cat, cols_cat, data_cat,
-                       x_rows):
+                       x_rows, col_items=False):
         rows = np.concatenate(rows_cat)
         cols = np.concatenate(cols_cat)
         data = np.concatenate(data_cat)
@@ -368,7 +377,8 @@ class ArrowSlimMPI(ArrowMatrix):
         np.cumsum(np.bincount(rows, minlength=n_rows), out=indptr[1:])
         handle = self.backend.upload_arrays((n_rows, n_cols), indptr,
                                             cols.astype(np.int32),
-                                            data.astype(np.float32))
+                                            data.astype(np.float32),
+                                            col_items=col_items)
         handle.x_rows = x_rows  # algorithmic X rows for the roofline
         return handle
 

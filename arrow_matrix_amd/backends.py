@@ -106,9 +106,10 @@ class GpuBackend:
         assert X.is_contiguous() and C.is_contiguous()
         block.spmm(X.data_ptr(), C.data_ptr(), C.shape[1], beta, self._stream())
 
-    def upload_arrays(self, shape, indptr, indices, data, row_ids=None) -> hip.CsrBlockGPU:
+    def upload_arrays(self, shape, indptr, indices, data, row_ids=None,
+                      col_items=False) -> hip.CsrBlockGPU:
         return hip.CsrBlockGPU(arrays=(shape, indptr, indices, data),
-                               row_ids=row_ids)
+                               row_ids=row_ids, col_items=col_items)
 
     def spmm_dual(self, block: hip.CsrBlockGPU, X0: torch.Tensor,
                   X1: torch.Tensor, C: torch.Tensor, beta: int):

@@ -77,10 +77,10 @@ struct CsrBlock {
 };
 
 int env_queue_default() {
-  // -1 unset (per-launch policy: queue when GROUP >= 16 — measured
-  // +54 % at k=128/cfg4, +18 % at k=64, but -19 % at k=16 where the
-  // per-chunk barrier dominates 4-lane groups; profiles/
-  // r01_queue_chunk_sweep.txt, r01_ksweep_queue_20M.txt);
+  // -1 unset (per-launch policy: queue when GROUP >= 8 — measured
+  // +54 % at k=128/cfg4, +18 % at k=64, +2 % at k=32 same-box, but
+  // -19 % at k=16 where the per-chunk barrier dominates 4-lane groups;
+  // profiles/r01_queue_chunk_sweep.txt, r01_ksweep_queue_20M.txt);
   // ARROW_QUEUE=0 forces grid-stride, ARROW_QUEUE=1 forces queues.
   static const int v = [] {
     const char *e = getenv("ARROW_QUEUE");
@@ -565,7 +565,7 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
     return (e && e[0] == '1') ? 1 : 0;
   }();
   const int qd = blk.queue_mode >= 0 ? blk.queue_mode : env_queue_default();
-  const bool useq = blk.qseg && (qd >= 0 ? qd : (GROUP >= 16));
+  const bool useq = blk.qseg && (qd >= 0 ? qd : (GROUP >= 8));
   // Queue mode: size the grid to residency (8 blocks/CU fit at this
   // occupancy — 4 waves/WG, 8 waves/SIMD), not to the item count; chunk =
   // 2 rounds per grab keeps each XCD's in-flight row window a few MB.
@@ -662,7 +662,8 @@ int arrow_synchronize(void) {
 
 static int64_t csr_create_impl(int64_t rows, int64_t cols, int64_t nnz,
                                const int64_t *indptr, const int32_t *indices,
-                               const float *data, const int64_t *row_ids) {
+                               const float *data, const int64_t *row_ids,
+                               int flags = 0) {
   if (rows < 0 || cols < 0 || nnz < 0 || (rows > 0 && !indptr)) {
     set_error("arrow_csr_create: bad arguments");
     return -1;
@@ -701,8 +702,40 @@ static int64_t csr_create_impl(int64_t rows, int64_t cols, int64_t nnz,
   blk.n_items = (int64_t)item_row.size();
   blk.n_split_rows = (int64_t)split_rows.size();
 
+  // ARROW_CSR_COL_ITEMS: order work items by their first column instead of
+  // by row. For hub-heavy structures (the first block-row: long CSR rows
+  // split into column-contiguous items) this makes a queue segment a
+  // COLUMN window of X, so the ~concurrent hub-row streams share the
+  // XCD's L2. Output correctness is order-independent (each non-split row
+  // appears once; split rows accumulate atomically either way).
+  if (flags & 1) {
+    const int64_t n_it = (int64_t)item_row.size();
+    std::vector<int64_t> order(n_it);
+    for (int64_t i = 0; i < n_it; ++i) order[i] = i;
+    std::stable_sort(order.begin(), order.end(),
+                     [&](int64_t a, int64_t b) {
+                       const int32_t ca = item_begin[a] < item_end[a]
+                                              ? indices[item_begin[a]]
+                                              : INT32_MAX;
+                       const int32_t cb = item_begin[b] < item_end[b]
+                                              ? indices[item_begin[b]]
+                                              : INT32_MAX;
+                       return ca < cb;
+                     });
+    std::vector<int32_t> r2(n_it), b2(n_it), e2(n_it);
+    for (int64_t i = 0; i < n_it; ++i) {
+      r2[i] = item_row[order[i]];
+      b2[i] = item_begin[order[i]];
+      e2[i] = item_end[order[i]];
+    }
+    item_row.swap(r2);
+    item_begin.swap(b2);
+    item_end.swap(e2);
+  }
+
   // nnz-balanced contiguous item segments for the per-XCD queue scheduler
-  // (items are in row order, so a segment is a contiguous row window)
+  // (items are in row order — or column order with flags&1 — so a segment
+  // is a contiguous window of X)
   int64_t qseg_h[9];
   {
     // weight = nnz + 1 per item: balances by data volume while keeping
@@ -758,6 +791,14 @@ int64_t arrow_csr_create_rows(int64_t rows, int64_t cols, int64_t nnz,
                               const int64_t *indptr, const int32_t *indices,
                               const float *data, const int64_t *row_ids) {
   return csr_create_impl(rows, cols, nnz, indptr, indices, data, row_ids);
+}
+
+int64_t arrow_csr_create_opts(int64_t rows, int64_t cols, int64_t nnz,
+                              const int64_t *indptr, const int32_t *indices,
+                              const float *data, const int64_t *row_ids,
+                              int flags) {
+  return csr_create_impl(rows, cols, nnz, indptr, indices, data, row_ids,
+                         flags);
 }
 
 int arrow_csr_destroy(int64_t handle) {

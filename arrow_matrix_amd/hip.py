@@ -45,6 +45,12 @@ def _load():
         ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
         ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
         ctypes.POINTER(ctypes.c_float), ctypes.POINTER(ctypes.c_int64)]
+    lib.arrow_csr_create_opts.restype = ctypes.c_int64
+    lib.arrow_csr_create_opts.argtypes = [
+        ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
+        ctypes.POINTER(ctypes.c_float), ctypes.POINTER(ctypes.c_int64),
+        ctypes.c_int]
     lib.arrow_csr_destroy.argtypes = [ctypes.c_int64]
     lib.arrow_csr_destroy.restype = ctypes.c_int
     lib.arrow_csr_nnz.argtypes = [ctypes.c_int64]
@@ -90,11 +96,12 @@ class CsrBlockGPU:
     """A CSR block resident in HBM (uploaded ONCE — unlike the reference,
     which re-uploads A every iteration, arrow_slim_mpi.py:184-232)."""
 
-    def __init__(self, csr=None, arrays=None, row_ids=None):
+    def __init__(self, csr=None, arrays=None, row_ids=None, col_items=False):
         """csr: scipy CSR, or arrays=(shape, indptr, indices, data) for raw
         uploads (the fused layouts use negative column indices, which scipy
         would reject). row_ids: optional explicit output-row id per
-        structure row (reordered layouts)."""
+        structure row (reordered layouts). col_items: order work items by
+        first column (hub structures; see arrow_csr_create_opts)."""
         lib = _load()
         if arrays is not None:
             (rows, cols), indptr, indices, data = arrays
@@ -107,7 +114,18 @@ class CsrBlockGPU:
         data = np.ascontiguousarray(data, dtype=np.float32)
         self.shape = (rows, cols)
         self.nnz = int(indices.size)
-        if row_ids is not None:
+        if col_items:
+            rid = (np.ascontiguousarray(row_ids, dtype=np.int64)
+                   if row_ids is not None else None)
+            self._handle = _check(lib.arrow_csr_create_opts(
+                rows, cols, self.nnz,
+                indptr.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+                indices.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+                data.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
+                (rid.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))
+                 if rid is not None else None), 1),
+                "arrow_csr_create_opts")
+        elif row_ids is not None:
             row_ids = np.ascontiguousarray(row_ids, dtype=np.int64)
             self._handle = _check(lib.arrow_csr_create_rows(
                 rows, cols, self.nnz,

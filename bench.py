@@ -158,7 +158,7 @@ def _kernel_name(k):
     group = 1
     while group < lanes and group < 64:
         group <<= 1
-    return "spmm_kernel_q" if group >= 16 else "spmm_kernel"
+    return "spmm_kernel_q" if group >= 8 else "spmm_kernel"
 
 
 def cpu_baseline_sample(w, band, k, threads):

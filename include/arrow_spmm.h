@@ -52,6 +52,15 @@ int arrow_abi_version(void);
 int64_t arrow_csr_create(int64_t rows, int64_t cols, int64_t nnz,
                          const int64_t *indptr, const int32_t *indices,
                          const float *data);
+/* As arrow_csr_create_rows plus option flags. flags bit 0: order work
+ * items by first column instead of row — for hub-heavy structures whose
+ * long rows split into column-contiguous items, a queue segment then
+ * covers a COLUMN window of X (locality for the per-XCD scheduler).
+ * row_ids may be NULL. */
+int64_t arrow_csr_create_opts(int64_t rows, int64_t cols, int64_t nnz,
+                              const int64_t *indptr, const int32_t *indices,
+                              const float *data, const int64_t *row_ids,
+                              int flags);
 /* As arrow_csr_create, but structure row r writes output row row_ids[r]
  * (reordered layouts, e.g. hub-sorted first-block-column entries). */
 int64_t arrow_csr_create_rows(int64_t rows, int64_t cols, int64_t nnz,

@@ -240,3 +240,58 @@ def test_spmm_queue_dual_negative_columns(gpu):
             xrow = X1[-c - 1] if c < 0 else X0[c]
             ref[r] += float(vals[t]) * xrow
     np.testing.assert_allclose(Ct.cpu().numpy(), ref, rtol=2e-5, atol=2e-5)
+
+
+def test_spmm_col_items_hub(gpu):
+    """Column-ordered work items (arrow_csr_create_opts flags=1): identical
+    results on a hub-heavy structure with split rows and empty rows."""
+    from arrow_matrix_amd import hip
+    rows, cols, k = 300, 4000, 128
+    rs = np.random.RandomState(5)
+    A = sparse.random(rows, cols, density=0.003, format='lil', random_state=rs,
+                      dtype=np.float64)
+    A[3, :] = rs.rand(cols)
+    A[177, :] = rs.rand(cols)
+    A[50, :] = 0
+    A = sparse.csr_matrix(A, dtype=np.float32)
+    rng = np.random.default_rng(6)
+    X = (2 * rng.random((cols, k)) - 1).astype(np.float32)
+    Xt = torch.from_numpy(X).cuda()
+    for beta in (0, 1):
+        C0 = (2 * rng.random((rows, k)) - 1).astype(np.float32)
+        Ct = torch.from_numpy(C0.copy()).cuda()
+        blk = hip.CsrBlockGPU(A, col_items=True)
+        blk.set_queue(1)
+        blk.spmm(Xt.data_ptr(), Ct.data_ptr(), k, beta,
+                 torch.cuda.current_stream().cuda_stream)
+        torch.cuda.synchronize()
+        ref = (A @ X) + (C0 if beta else 0)
+        scale = max(1.0, float(np.abs(ref).max()))
+        np.testing.assert_allclose(Ct.cpu().numpy(), ref, rtol=2e-5,
+                                   atol=2e-5 * scale)
+
+
+def test_spmm_col_items_row_ids(gpu):
+    """col_items composed with explicit row_ids (reordered output rows)."""
+    from arrow_matrix_amd import hip
+    rng = np.random.default_rng(8)
+    n_struct, out_rows, cols, k = 50, 80, 600, 32
+    nnz_pr = 5
+    indptr = np.arange(0, (n_struct + 1) * nnz_pr, nnz_pr, dtype=np.int64)
+    idx = rng.integers(0, cols, n_struct * nnz_pr).astype(np.int32)
+    vals = (2 * rng.random(n_struct * nnz_pr) - 1).astype(np.float32)
+    row_ids = rng.choice(out_rows, n_struct, replace=False).astype(np.int64)
+    blk = hip.CsrBlockGPU(arrays=((n_struct, cols), indptr, idx, vals),
+                          row_ids=row_ids, col_items=True)
+    blk.set_queue(1)
+    X = (2 * rng.random((cols, k)) - 1).astype(np.float32)
+    Xt = torch.from_numpy(X).cuda()
+    Ct = torch.full((out_rows, k), 7.0, device='cuda')
+    blk.spmm(Xt.data_ptr(), Ct.data_ptr(), k, 1,
+             torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    ref = np.full((out_rows, k), 7.0, dtype=np.float64)
+    for r in range(n_struct):
+        for t in range(indptr[r], indptr[r + 1]):
+            ref[row_ids[r]] += float(vals[t]) * X[idx[t]]
+    np.testing.assert_allclose(Ct.cpu().numpy(), ref, rtol=2e-5, atol=2e-5)

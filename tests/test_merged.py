@@ -18,7 +18,7 @@ class _CaptureBackend:
     def __init__(self):
         self.uploads = []
 
-    def upload_arrays(self, shape, indptr, indices, data, row_ids=None):
+    def upload_arrays(self, shape, indptr, indices, data, row_ids=None, col_items=False):
         h = SimpleNamespace(shape=shape, indptr=np.asarray(indptr),
                             indices=np.asarray(indices),
                             data=np.asarray(data), nnz=int(len(indices)),
