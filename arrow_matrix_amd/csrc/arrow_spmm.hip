@@ -19,8 +19,11 @@
 //     item, so beta=0 also zeroes empty rows.
 //   * A's (col, val) stream is read once per item by all lanes of the
 //     group (same-address broadcast within the wave's transaction).
-//   * Work items are consumed via a grid-stride loop with >= 4 items per
-//     workgroup so the 256-CU / 8-XCD chip is filled for any block shape.
+//   * Work items are consumed either by a grid-stride loop (small GROUP)
+//     or by the per-XCD queue scheduler (GROUP >= 8 by default): 8
+//     contiguous nnz-balanced item segments drained via per-XCD atomic
+//     chunk counters so each XCD walks one tight X window in order
+//     (see spmm_kernel_q below; measured +54 % at k=128).
 
 #include <hip/hip_runtime.h>
 #include <algorithm>
