@@ -562,10 +562,13 @@ template <int VEC, int GROUP>
 int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
                    float *C, int64_t k, int beta, hipStream_t stream) {
   constexpr int GROUPS_PER_BLOCK = BLOCK_THREADS / GROUP;
-  // A/B: non-temporal pairs load + C store (measured ~0 at default shapes)
+  // Non-temporal pairs load + C store: default ON (+1.4-1.7 % at cfg4
+  // with the queue scheduler — the once-read A stream and once-written C
+  // stop evicting the X window; profiles/r01_nt_chunk_k32_ab.txt,
+  // r01_colsort_nt_ab.txt). ARROW_SPMM_NT=0 restores cached accesses.
   static const int nt_mode = [] {
     const char *e = getenv("ARROW_SPMM_NT");
-    return (e && e[0] == '1') ? 1 : 0;
+    return (e && e[0] == '0') ? 0 : 1;
   }();
   const int qd = blk.queue_mode >= 0 ? blk.queue_mode : env_queue_default();
   const bool useq = blk.qseg && (qd >= 0 ? qd : (GROUP >= 8));
