@@ -340,6 +340,10 @@ def main():
     if use_gpu and events:
         total_ms = 0.0
         total_bytes = 0.0
+        if os.environ.get('ARROW_BENCH_LAUNCH_DETAIL') == '1':
+            for i, (s, e, nnz, c_rows, x_rows) in enumerate(events):
+                print(f"launch[{i}]: {s.elapsed_time(e):.3f} ms nnz={nnz} "
+                      f"c_rows={c_rows} x_rows={x_rows}", file=sys.stderr)
         for s, e, nnz, c_rows, x_rows in events:
             total_ms += s.elapsed_time(e)
             # algorithmic bytes: A pairs once + work items +
