@@ -571,7 +571,11 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
     return (e && e[0] == '0') ? 0 : 1;
   }();
   const int qd = blk.queue_mode >= 0 ? blk.queue_mode : env_queue_default();
-  const bool useq = blk.qseg && (qd >= 0 ? qd : (GROUP >= 8));
+  // policy default: queues need wide groups AND enough work per structure —
+  // below ~32M nnz the per-grab atomic+barrier overhead outweighs the L2
+  // window benefit (cfg3-size structures measured -20 % under queues)
+  const bool useq = blk.qseg &&
+      (qd >= 0 ? qd : (GROUP >= 8 && blk.nnz >= (32LL << 20)));
   // Queue mode: size the grid to residency (8 blocks/CU fit at this
   // occupancy — 4 waves/WG, 8 waves/SIMD), not to the item count; chunk =
   // 2 rounds per grab keeps each XCD's in-flight row window a few MB.

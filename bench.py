@@ -145,9 +145,10 @@ def build_blocks_for_rank(rank, world, w, nb, parts, gen_device, band):
     return grids, first, last
 
 
-def _kernel_name(k):
+def _kernel_name(k, nnz_dominant):
     """Mirror the launcher's scheduler policy (csrc/arrow_spmm.hip): queue
-    scheduler when GROUP >= 16 unless ARROW_QUEUE forces it."""
+    scheduler when GROUP >= 8 and the structure is big enough, unless
+    ARROW_QUEUE forces it. nnz_dominant: nnz of the dominant structure."""
     env = os.environ.get('ARROW_QUEUE', '')
     if env == '0':
         return "spmm_kernel"
@@ -158,7 +159,8 @@ def _kernel_name(k):
     group = 1
     while group < lanes and group < 64:
         group <<= 1
-    return "spmm_kernel_q" if group >= 8 else "spmm_kernel"
+    big = nnz_dominant >= (32 << 20)
+    return "spmm_kernel_q" if (group >= 8 and big) else "spmm_kernel"
 
 
 def cpu_baseline_sample(w, band, k, threads):
@@ -367,7 +369,7 @@ def main():
             "unit": "GB/s",
             "frac": round(achieved / peak, 4),
             "traffic": traffic,
-            "kernel": _kernel_name(k),
+            "kernel": _kernel_name(k, max(n for _, _, n, _, _ in events)),
             "launches": len(events),
             "avg_launch_ms": round(total_ms / len(events), 4),
         }
