@@ -30,7 +30,9 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def run_pass(counters, outdir, bench_args):
-    env = dict(os.environ, TMPDIR='/tmp')
+    # ARROW_TRAFFIC_CHILD stops the bench child from launching its own
+    # traffic probe (bench.py runs this script automatically at N=1)
+    env = dict(os.environ, TMPDIR='/tmp', ARROW_TRAFFIC_CHILD='1')
     cmd = ['rocprofv3', '--pmc', *counters, '--kernel-include-regex',
            'spmm_kernel', '-d', outdir, '--',
            sys.executable, os.path.join(REPO, 'bench.py'), *bench_args,
@@ -60,11 +62,14 @@ def run_pass(counters, outdir, bench_args):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument('--rows', default='20000000')
+    ap.add_argument('--features', default='128')
+    ap.add_argument('--band', default='1024')
     ap.add_argument('--steps', default='2')
     ap.add_argument('--warmup', default='1')
     ap.add_argument('--out', default=os.path.join(REPO, 'gpurun_out', 'traffic.json'))
     a = ap.parse_args()
-    bench_args = ['--rows', a.rows, '--steps', a.steps, '--warmup', a.warmup]
+    bench_args = ['--rows', a.rows, '--features', a.features, '--band', a.band,
+                  '--steps', a.steps, '--warmup', a.warmup]
     base = os.path.join(REPO, 'gpurun_out')
     fetch = run_pass(['FETCH_SIZE'], os.path.join(base, 'tr_fetch'), bench_args)
     write = run_pass(['WRITE_SIZE'], os.path.join(base, 'tr_write'), bench_args)
