@@ -396,6 +396,13 @@ class ArrowSlimMPI(ArrowMatrix):
             if np.dtype(dtype) != self.backend.np_dtype:
                 self.backend = make_backend(self.device, dtype)
                 self.C_i = self.C_0 = self.X_i = self.X_0 = None
+        if self.width is not None and self.nnz_owned:
+            # blocks were loaded at a fixed block size; a different width
+            # would silently mis-tile the resident structures against the
+            # X/C buffers — refuse instead
+            assert number_of_rows_per_rank == self.width, (
+                f"zero_rhs width {number_of_rows_per_rank} != block size "
+                f"{self.width} fixed at load time")
         self.width = number_of_rows_per_rank
         w, k = number_of_rows_per_rank, number_of_columns
         stripe = (max(self.n_owned, 1) * w, k)

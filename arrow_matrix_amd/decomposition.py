@@ -36,7 +36,13 @@ class ArrowGraph:
                  arrow_width: int):
         self.graph = matrix
         self.permutation = np.asarray(permutation, dtype=np.int64)
-        self.nonzero_rows = int(np.count_nonzero(matrix.getnnz(1)))
+        # NOTE: despite the name, the reference's same-named attribute counts
+        # ZERO-degree vertices (decomposition.py:20: `len([d for d in
+        # graph.degree() if d == 0])`) and persists that to _nnzrows.npy
+        # (graphio.py:189-191); kept bit-compatible here. degree = row+col
+        # nonzeros (igraph counts in+out for directed graphs).
+        deg = matrix.getnnz(1) + matrix.getnnz(0)
+        self.nonzero_rows = int(np.count_nonzero(deg == 0))
         self.arrow_width = int(arrow_width)
 
     def __getitem__(self, item):

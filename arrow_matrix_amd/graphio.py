@@ -69,19 +69,39 @@ def save_decomposition_new(decomposition, filename: str, width: int,
                            block_diagonal: bool = True) -> None:
     """Save parts as *_indptr/_indices/_data/_permutation .npy files.
 
-    `decomposition` is a list of (csr_matrix, permutation) pairs; `width` is
-    the decomposition's arrow width (shared across parts, as the reference
-    producer guarantees). File layout identical to reference
-    graphio.py:171-191 (the graph pickle / adjacency of the `save_graph`
-    branch is intentionally not written; the hot path never reads it).
+    `decomposition` is a list of (csr_matrix, permutation) pairs or
+    ArrowGraph-like objects. Each part is saved under its OWN arrow width
+    when the entry carries one (`.arrow_width`), exactly as the reference
+    does (graphio.py:176-183: the last level can report a larger actual
+    width); plain pairs use the caller's `width`. The per-part zero-degree
+    counts are persisted to `_nnzrows.npy` under part 0's width
+    (reference graphio.py:189-191; the attribute is named `nonzero_rows`
+    but counts zero-degree vertices — see decomposition.ArrowGraph).
+    File layout identical to reference graphio.py:171-191 (the graph
+    pickle / adjacency of the `save_graph` branch is intentionally not
+    written; the hot path never reads it).
     """
-    for i, (B, permutation) in enumerate(decomposition):
+    widths = []
+    zero_deg_counts = []
+    for i, entry in enumerate(decomposition):
+        B, permutation = entry[0], entry[1]
+        w_i = int(getattr(entry, 'arrow_width', width))
+        widths.append(w_i)
         B = sparse.csr_matrix(B)
-        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.indptr_npy), B.indptr)
-        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.indices_npy), B.indices)
-        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.data_npy), B.data)
-        np.save(format_path(filename, width, i, block_diagonal, DecompositionFileType.permutation_npy),
+        if hasattr(entry, 'nonzero_rows'):
+            zero_deg_counts.append(int(entry.nonzero_rows))
+        else:
+            deg = B.getnnz(1) + B.getnnz(0)
+            zero_deg_counts.append(int(np.count_nonzero(deg == 0)))
+        np.save(format_path(filename, w_i, i, block_diagonal, DecompositionFileType.indptr_npy), B.indptr)
+        np.save(format_path(filename, w_i, i, block_diagonal, DecompositionFileType.indices_npy), B.indices)
+        np.save(format_path(filename, w_i, i, block_diagonal, DecompositionFileType.data_npy), B.data)
+        np.save(format_path(filename, w_i, i, block_diagonal, DecompositionFileType.permutation_npy),
                 np.asarray(permutation))
+    if widths:
+        np.save(format_path(filename, widths[0], 0, block_diagonal,
+                            DecompositionFileType.nonzero_rows_npy),
+                np.asarray(zero_deg_counts, dtype=np.int64))
 
 
 def load_decomposition_new(filename: str, width: Optional[int] = None,
@@ -116,6 +136,22 @@ def load_decomposition_new(filename: str, width: Optional[int] = None,
                 f = format_path(filename, width, i, block_diagonal, DecompositionFileType.permutation_npy)
                 permutation = np.load(f)
         except FileNotFoundError:
+            # A part saved under a DIFFERENT width suffix (the reference
+            # saves each part under its own arrow_width, graphio.py:176-183)
+            # would be silently truncated here — warn instead of hiding it.
+            import glob as _glob
+            import warnings
+            bd = "_bd" if block_diagonal else ""
+            pattern = f"{filename}_B_*_{i}{bd}_indptr.npy"
+            others = [p for p in _glob.glob(pattern)
+                      if p != format_path(filename, width, i, block_diagonal,
+                                          DecompositionFileType.indptr_npy)]
+            if others:
+                warnings.warn(
+                    f"load_decomposition_new: stopping at part {i} for "
+                    f"width={width}, but that part exists under a different "
+                    f"width suffix: {sorted(others)} (per-part arrow widths "
+                    f"— load with the matching width)")
             break
         decomposition.append((B, permutation))
         i += 1

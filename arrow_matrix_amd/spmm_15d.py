@@ -58,6 +58,7 @@ class Spmm15D:
         self.backend = make_backend(device)
         A = sparse.csr_matrix(A)
         NI, NK = A.shape
+        self.NK = NK
         self.NJ = X_cols
         self.lNI = int(np.ceil(NI / p_div_c))
         self.lNKb = int(np.ceil(NK / p_div_c))
@@ -97,8 +98,13 @@ class Spmm15D:
                     self.reduce_comm = TorchDistComm(g)
 
     def x_block_rows(self) -> int:
-        """Rows of this rank's X block (bcast rank = x coordinate)."""
-        return self.block_cols[0] if self.rounds else 0
+        """Rows of the X block this rank broadcasts (block index = x
+        coordinate): min(NK, (x+1)*lNKb) - x*lNKb, which differs from the
+        rank's own first A sub-block width on ragged NK."""
+        if not self.rounds:
+            return 0
+        lo = min(self.NK, self.x * self.lNKb)
+        return min(self.NK, (self.x + 1) * self.lNKb) - lo
 
     def spmm(self, X_local: torch.Tensor) -> torch.Tensor:
         """One Y = A @ X round sweep (reference spmm_15d_cpu :312-370).
