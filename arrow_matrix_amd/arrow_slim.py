@@ -177,12 +177,16 @@ class ArrowSlimMPI(ArrowMatrix):
         return n_chunks
 
     @property
-    def _row0_col_items(self) -> bool:
-        """Order the row-0 (hub) structures' work items by column so a
-        per-XCD queue segment covers a column window of X (A/B knob;
-        see arrow_csr_create_opts in include/arrow_spmm.h)."""
+    def _row0_col_items(self) -> int:
+        """Work-item ordering of the row-0 (hub) structures (A/B knob; see
+        arrow_csr_create_opts in include/arrow_spmm.h): 0 row order,
+        1 global column sort (a queue segment becomes a column window of
+        X), 2 column sort WITHIN each nnz-balanced row segment (two-level:
+        each XCD keeps its contiguous C_0 range but walks its X gathers in
+        column order)."""
         import os as _os
-        return _os.environ.get('ARROW_ROW0_COLSORT', '0') == '1'
+        v = _os.environ.get('ARROW_ROW0_COLSORT', '0')
+        return int(v) if v in ('0', '1', '2') else 0
 
     def _build_merged_gpu(self) -> None:
         w = self.width
@@ -225,6 +229,7 @@ class ArrowSlimMPI(ArrowMatrix):
             data = np.concatenate(data_cat)
             bounds = [w * q // n_chunks for q in range(n_chunks + 1)]
             self._A_row0 = []
+            qb_row0 = int(_os.environ.get('ARROW_Q_BLOCKS_ROW0', '0'))
             for q in range(n_chunks):
                 lo, hi = bounds[q], bounds[q + 1]
                 m = (rows >= lo) & (rows < hi)
@@ -232,6 +237,8 @@ class ArrowSlimMPI(ArrowMatrix):
                                         [rows[m] - lo], [cols[m]], [data[m]],
                                         x_rows=nw * w // n_chunks,
                                         col_items=self._row0_col_items)
+                if qb_row0:
+                    h.set_qblocks(qb_row0)
                 self._A_row0.append((h, lo, hi))
         # --- rest merge: C[r] = A_rr @ X_r + A_r0 @ X_0
         #     (+ interior banded off-diagonals A_{r,r±1} @ X_{r±1} when the
@@ -302,6 +309,10 @@ class ArrowSlimMPI(ArrowMatrix):
             self._A_rest = self._merged_chunks(rest_rows, nw * w, rows_cat,
                                                cols_cat, data_cat,
                                                x_rows_total, cap)
+            qb_rest = int(_os.environ.get('ARROW_Q_BLOCKS_REST', '0'))
+            if qb_rest:
+                for h, _, _ in self._A_rest:
+                    h.set_qblocks(qb_rest)
             if split_col and col_rows:
                 self._A_col = self._merged_col_sorted(rest_rows, nw * w,
                                                       col_rows, col_cols,
@@ -548,6 +559,24 @@ class ArrowSlimMPI(ArrowMatrix):
         # arrow_slim_mpi.py:181-195); each chunk's (all)reduce starts as
         # soon as its slice is computed and pipelines behind the remaining
         # compute (arrow_slim_mpi.py:116's single blocking Reduce).
+        #
+        # ARROW_PAR_ROW0=1: launch the row-0 (hub) structure on a SIDE HIP
+        # stream so it runs CONCURRENTLY with the rest launch below — the
+        # hub gathers are fabric-bound while the banded rest launch is not,
+        # so co-running them fills the slack. Fork-join via
+        # events (capture-legal for hipGraph).
+        import os as _os
+        par = (_os.environ.get('ARROW_PAR_ROW0', '0') == '1'
+               and bool(self._A_row0))
+        stream_ctx = None
+        if par:
+            if not hasattr(self, '_row0_stream') or self._row0_stream is None:
+                self._row0_stream = torch.cuda.Stream()
+            ev_x = torch.cuda.Event()
+            ev_x.record()  # X_i (and X_0 copy) ready on the main stream
+            self._row0_stream.wait_event(ev_x)
+            stream_ctx = torch.cuda.stream(self._row0_stream)
+            stream_ctx.__enter__()
         tic = time.perf_counter()
         reduce_works = []
         if self._A_row0:
@@ -578,6 +607,10 @@ class ArrowSlimMPI(ArrowMatrix):
                 else:
                     reduce_works.append(
                         self.comm.reduce_sum_(C_sl, dst=0, async_op=True))
+        if par:
+            ev_r = torch.cuda.Event()
+            ev_r.record(self._row0_stream)
+            stream_ctx.__exit__(None, None, None)
         wb_logging.log({"spmm_row_reduce": time.perf_counter() - tic})
 
         # C_rest = A_diag_merged @ X_stripe + A_col_merged @ X_0 fused:
@@ -606,6 +639,10 @@ class ArrowSlimMPI(ArrowMatrix):
                 Cr = self.C_i[local:local + w]
                 self._timed(lambda: be.spmm_block(hdl, halo, Cr, 1),
                             hdl.nnz, w, w)
+        if par:
+            # join: the main stream's C_0 consumers (head copy / X_0 swap)
+            # wait for the side-stream row-0 work
+            torch.cuda.current_stream().wait_event(ev_r)
         for wk in reduce_works:
             if wk is not None:
                 wk.wait()

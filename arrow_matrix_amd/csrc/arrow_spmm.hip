@@ -77,6 +77,7 @@ struct CsrBlock {
   int queue_mode = -1;
   int64_t *qseg = nullptr;  // device, 9 nnz-balanced item-segment bounds
   int32_t *qctr = nullptr;  // device, 8 chunk counters padded 32 ints apart
+  int q_blocks = 0;         // per-structure grid override (0 = ARROW_Q_BLOCKS)
 };
 
 int env_queue_default() {
@@ -408,6 +409,67 @@ __global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel_q(
   }
 }
 
+// Per-WAVE queue grab variant of spmm_kernel_q: each 64-lane wave pulls its
+// own chunk from the XCD's counter (lane-0 atomicAdd broadcast by wave
+// shuffle) — no __syncthreads per grab. Removes the whole-block rendezvous
+// that made queues lose at small GROUP (k=16: 16 groups per wave stall on
+// the slowest group of all 4 waves), at the cost of 4x the atomic rate
+// (fine: per-XCD counters, dequeue ~0.25-1.1 us uncontended and these are
+// thousands of grabs per segment, MI355X_MICROARCH.md §dequeue).
+template <int VEC, int GROUP, int BETA, bool GUARD>
+__global__ __launch_bounds__(BLOCK_THREADS) void spmm_kernel_qw(
+    const int2 *__restrict__ pairs,
+    const int32_t *__restrict__ item_row, const int32_t *__restrict__ item_begin,
+    const int32_t *__restrict__ item_end,
+    const int64_t *__restrict__ qseg,  // 9 segment bounds (items)
+    int32_t *__restrict__ qctr,       // 8 counters, padded 32 ints apart
+    int chunk_items,                  // per-WAVE grab size (items)
+    const float *__restrict__ X0, const float *__restrict__ X1,
+    float *__restrict__ C, int64_t k, int64_t col_off, int nt_mode) {
+  constexpr int GROUPS_PER_WAVE = 64 / GROUP;
+  const int lane = threadIdx.x & 63;
+  const int lane_in_group = threadIdx.x % GROUP;
+  const int group_in_wave = lane / GROUP;
+  const int64_t col0 = col_off + (int64_t)lane_in_group * VEC;
+  const bool active = !GUARD || (col0 < k);
+
+  const unsigned q0 = arrow_xcc_id() & 7;
+  for (unsigned qi = 0; qi < 8; ++qi) {
+    const unsigned q = (q0 + qi) & 7;
+    const int64_t lo = qseg[q], hi = qseg[q + 1];
+    if (lo >= hi) continue;
+    for (;;) {
+      int grab = 0;
+      if (lane == 0) grab = atomicAdd(&qctr[q * 32], chunk_items);
+      grab = __shfl(grab, 0, 64);
+      const int64_t base = lo + grab;
+      if (base >= hi) break;
+      const int64_t end = base + chunk_items < hi ? base + chunk_items : hi;
+      int64_t item = base + group_in_wave;
+      int32_t next_row = 0, next_b = 0, next_e = 0;
+      if (item < end) {
+        next_row = __builtin_nontemporal_load(&item_row[item]);
+        next_b = __builtin_nontemporal_load(&item_begin[item]);
+        next_e = __builtin_nontemporal_load(&item_end[item]);
+      }
+      for (; item < end; item += GROUPS_PER_WAVE) {
+        const int32_t row_raw = next_row;
+        const int32_t b = next_b;
+        const int32_t e = next_e;
+        const int64_t nxt = item + GROUPS_PER_WAVE;
+        if (nxt < end) {
+          next_row = __builtin_nontemporal_load(&item_row[nxt]);
+          next_b = __builtin_nontemporal_load(&item_begin[nxt]);
+          next_e = __builtin_nontemporal_load(&item_end[nxt]);
+        }
+        spmm_process_item<VEC, GROUP, BETA, GUARD>(
+            pairs, row_raw, b, e, X0, X1, C, k, col0, active, lane_in_group,
+            nt_mode);
+      }
+    }
+  }
+}
+
 __global__ void zero_rows_kernel(float *__restrict__ C,
                                  const int32_t *__restrict__ rows,
                                  int64_t n_rows, int64_t k) {
@@ -571,11 +633,23 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
     return (e && e[0] == '0') ? 0 : 1;
   }();
   const int qd = blk.queue_mode >= 0 ? blk.queue_mode : env_queue_default();
+  // ARROW_QWAVE=1: per-wave queue grabs (spmm_kernel_qw) — no per-chunk
+  // block barrier, so the queue scheduler also pays at GROUP=4 (k=16)
+  static const int qwave = [] {
+    const char *e = getenv("ARROW_QWAVE");
+    return (e && e[0] == '1') ? 1 : 0;
+  }();
   // policy default: queues need wide groups AND enough work per structure —
-  // below ~32M nnz the per-grab atomic+barrier overhead outweighs the L2
-  // window benefit (cfg3-size structures measured -20 % under queues)
+  // below ~32M nnz (ARROW_Q_MIN_NNZ overrides) the per-grab overhead
+  // outweighs the L2 window benefit; the wave variant extends the policy
+  // to GROUP >= 4
+  static const int64_t q_min_nnz = [] {
+    const char *e = getenv("ARROW_Q_MIN_NNZ");
+    return e ? (int64_t)atoll(e) : (32LL << 20);
+  }();
   const bool useq = blk.qseg &&
-      (qd >= 0 ? qd : (GROUP >= 8 && blk.nnz >= (32LL << 20)));
+      (qd >= 0 ? qd : ((GROUP >= 8 || (qwave && GROUP >= 4)) &&
+                       blk.nnz >= q_min_nnz));
   // Queue mode: size the grid to residency (8 blocks/CU fit at this
   // occupancy — 4 waves/WG, 8 waves/SIMD), not to the item count; chunk =
   // 2 rounds per grab keeps each XCD's in-flight row window a few MB.
@@ -583,11 +657,15 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
     const char *e = getenv("ARROW_Q_CHUNK");
     return e ? std::max(1, atoi(e)) : 2;
   }();
-  static const int q_blocks = [] {
+  static const int q_blocks_env = [] {
     const char *e = getenv("ARROW_Q_BLOCKS");
     return e ? std::max(8, atoi(e)) : 2048;
   }();
+  const int q_blocks = blk.q_blocks > 0 ? blk.q_blocks : q_blocks_env;
   const int chunk_items = GROUPS_PER_BLOCK * q_chunk_mult;
+  // per-wave grabs pull GROUPS_PER_WAVE-item rounds; same default round
+  // count per grab as the block variant
+  const int chunk_items_w = (64 / GROUP) * q_chunk_mult;
   int blocks = (int)std::min<int64_t>(
       (blk.n_items + GROUPS_PER_BLOCK - 1) / GROUPS_PER_BLOCK,
       useq ? q_blocks : 8192);
@@ -601,20 +679,28 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
                          blk.item_end, blk.n_items, X0, X1, C, k, col_off,
                          blk.xcd_remap, nt_mode);
     };
-    auto runq = [&](auto kern) {
+    auto runq = [&](auto kern, int chunk) {
       hipLaunchKernelGGL(kern, dim3(blocks), dim3(BLOCK_THREADS), 0, stream,
                          blk.pairs, blk.item_row, blk.item_begin,
-                         blk.item_end, blk.qseg, blk.qctr, chunk_items,
+                         blk.item_end, blk.qseg, blk.qctr, chunk,
                          X0, X1, C, k, col_off, nt_mode);
     };
     if (useq) {
       HIP_CHECK(hipMemsetAsync(blk.qctr, 0, 8 * 32 * sizeof(int32_t), stream));
-      if (beta == 0) {
-        if (guard) runq(spmm_kernel_q<VEC, GROUP, 0, true>);
-        else       runq(spmm_kernel_q<VEC, GROUP, 0, false>);
+      if (qwave) {
+        if (beta == 0) {
+          if (guard) runq(spmm_kernel_qw<VEC, GROUP, 0, true>, chunk_items_w);
+          else       runq(spmm_kernel_qw<VEC, GROUP, 0, false>, chunk_items_w);
+        } else {
+          if (guard) runq(spmm_kernel_qw<VEC, GROUP, 1, true>, chunk_items_w);
+          else       runq(spmm_kernel_qw<VEC, GROUP, 1, false>, chunk_items_w);
+        }
+      } else if (beta == 0) {
+        if (guard) runq(spmm_kernel_q<VEC, GROUP, 0, true>, chunk_items);
+        else       runq(spmm_kernel_q<VEC, GROUP, 0, false>, chunk_items);
       } else {
-        if (guard) runq(spmm_kernel_q<VEC, GROUP, 1, true>);
-        else       runq(spmm_kernel_q<VEC, GROUP, 1, false>);
+        if (guard) runq(spmm_kernel_q<VEC, GROUP, 1, true>, chunk_items);
+        else       runq(spmm_kernel_q<VEC, GROUP, 1, false>, chunk_items);
       }
     } else if (beta == 0) {
       if (guard) run(spmm_kernel<VEC, GROUP, 0, true>);
@@ -763,6 +849,41 @@ static int64_t csr_create_impl(int64_t rows, int64_t cols, int64_t nnz,
     while (target_q <= 8) qseg_h[target_q++] = n_it;
   }
 
+  // flags&2: TWO-LEVEL hub ordering — keep the 8 nnz-balanced ROW segments
+  // (so each XCD still owns a contiguous C range and load stays balanced),
+  // then sort the items WITHIN each segment by first column: an XCD's
+  // queue walk becomes a column-window sweep of X while its C writes stay
+  // inside the segment's row range. Order-independent for correctness
+  // (each non-split row appears once; split rows accumulate atomically).
+  if (flags & 2) {
+    for (int q = 0; q < 8; ++q) {
+      const int64_t lo = qseg_h[q], hi = qseg_h[q + 1];
+      if (hi - lo < 2) continue;
+      std::vector<int64_t> order((size_t)(hi - lo));
+      for (int64_t i = lo; i < hi; ++i) order[(size_t)(i - lo)] = i;
+      std::stable_sort(order.begin(), order.end(),
+                       [&](int64_t a, int64_t b) {
+                         const int32_t ca = item_begin[(size_t)a] < item_end[(size_t)a]
+                                                ? indices[item_begin[(size_t)a]]
+                                                : INT32_MAX;
+                         const int32_t cb = item_begin[(size_t)b] < item_end[(size_t)b]
+                                                ? indices[item_begin[(size_t)b]]
+                                                : INT32_MAX;
+                         return ca < cb;
+                       });
+      std::vector<int32_t> r2((size_t)(hi - lo)), b2((size_t)(hi - lo)),
+          e2((size_t)(hi - lo));
+      for (int64_t i = 0; i < hi - lo; ++i) {
+        r2[(size_t)i] = item_row[(size_t)order[(size_t)i]];
+        b2[(size_t)i] = item_begin[(size_t)order[(size_t)i]];
+        e2[(size_t)i] = item_end[(size_t)order[(size_t)i]];
+      }
+      std::copy(r2.begin(), r2.end(), item_row.begin() + (size_t)lo);
+      std::copy(b2.begin(), b2.end(), item_begin.begin() + (size_t)lo);
+      std::copy(e2.begin(), e2.end(), item_end.begin() + (size_t)lo);
+    }
+  }
+
   // pack (col, val) into 8-byte pairs for the staged kernel loads
   std::vector<int2> pairs((size_t)nnz);
   for (int64_t t = 0; t < nnz; ++t) {
@@ -850,6 +971,16 @@ int arrow_csr_set_queue(int64_t handle, int mode) {
     return -1;
   }
   it->second.queue_mode = mode < 0 ? -1 : (mode ? 1 : 0);
+  return 0;
+}
+
+int arrow_csr_set_qblocks(int64_t handle, int blocks) {
+  auto it = g_blocks.find(handle);
+  if (it == g_blocks.end()) {
+    set_error("arrow_csr_set_qblocks: bad handle");
+    return -1;
+  }
+  it->second.q_blocks = blocks > 0 ? blocks : 0;
   return 0;
 }
 

@@ -59,6 +59,8 @@ def _load():
     lib.arrow_csr_set_xcd_remap.restype = ctypes.c_int
     lib.arrow_csr_set_queue.argtypes = [ctypes.c_int64, ctypes.c_int]
     lib.arrow_csr_set_queue.restype = ctypes.c_int
+    lib.arrow_csr_set_qblocks.argtypes = [ctypes.c_int64, ctypes.c_int]
+    lib.arrow_csr_set_qblocks.restype = ctypes.c_int
     lib.arrow_spmm.argtypes = [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
                                ctypes.c_int64, ctypes.c_int, ctypes.c_void_p]
     lib.arrow_spmm.restype = ctypes.c_int
@@ -100,8 +102,10 @@ class CsrBlockGPU:
         """csr: scipy CSR, or arrays=(shape, indptr, indices, data) for raw
         uploads (the fused layouts use negative column indices, which scipy
         would reject). row_ids: optional explicit output-row id per
-        structure row (reordered layouts). col_items: order work items by
-        first column (hub structures; see arrow_csr_create_opts)."""
+        structure row (reordered layouts). col_items: work-item ordering
+        mode (hub structures; see arrow_csr_create_opts flags): 0/False
+        row order, 1/True global column sort, 2 column sort WITHIN each
+        XCD queue segment (two-level)."""
         lib = _load()
         if arrays is not None:
             (rows, cols), indptr, indices, data = arrays
@@ -123,7 +127,7 @@ class CsrBlockGPU:
                 indices.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
                 data.ctypes.data_as(ctypes.POINTER(ctypes.c_float)),
                 (rid.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))
-                 if rid is not None else None), 1),
+                 if rid is not None else None), int(col_items)),
                 "arrow_csr_create_opts")
         elif row_ids is not None:
             row_ids = np.ascontiguousarray(row_ids, dtype=np.int64)
@@ -155,6 +159,11 @@ class CsrBlockGPU:
         """Per-XCD queue scheduler: 1 on, 0 off, -1 follow ARROW_QUEUE env."""
         _check(_load().arrow_csr_set_queue(self._handle, int(mode)),
                "arrow_csr_set_queue")
+
+    def set_qblocks(self, blocks: int):
+        """Per-structure queue-grid override (0 = ARROW_Q_BLOCKS default)."""
+        _check(_load().arrow_csr_set_qblocks(self._handle, int(blocks)),
+               "arrow_csr_set_qblocks")
 
     def spmm_dual(self, X0_ptr: int, X1_ptr: int, C_ptr: int, k: int,
                   beta: int, stream: int = 0):

@@ -77,6 +77,11 @@ int arrow_csr_set_xcd_remap(int64_t handle, int enable);
  * consumers of each X row in ONE XCD's L2). mode: 1 on, 0 off,
  * -1 follow the ARROW_QUEUE env default. */
 int arrow_csr_set_queue(int64_t handle, int mode);
+/* Per-structure launch-grid override for the queue scheduler (workgroup
+ * count; 0 = the ARROW_Q_BLOCKS env default). Used to co-schedule two
+ * concurrent structure launches on separate HIP streams so neither grid
+ * fills the whole chip. */
+int arrow_csr_set_qblocks(int64_t handle, int blocks);
 
 /* C (+)= A @ X.  X: (cols, k) fp32 row-major device;  C: (rows, k).
  * beta = 0: C = A@X (rows not touched by A are zeroed);  beta = 1: C += A@X. */
