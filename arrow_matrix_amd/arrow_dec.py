@@ -64,6 +64,15 @@ class ArrowDecompositionMPI:
         self._to_next = to_next
         self._forward: List[Optional[_Exchange]] = []
         self._backward: List[Optional[_Exchange]] = []
+        # folded-permutation mode (single process, L > 1): parts i >= 1 are
+        # re-indexed through the composed forward/backward permutations into
+        # part 0's X/C numbering and run as extra beta=1 SpMM launches — the
+        # per-iteration permutation exchange disappears entirely (see
+        # _build_folded). _fold_maps[i] = (M_i, R_i): part-i X row r reads
+        # part-0 X row M_i[r]; part-i C row r accumulates into part-0 C row
+        # R_i[r] (-1 = unmapped).
+        self._folded: Optional[List] = None
+        self._fold_maps: Optional[List] = None
         self._initialize_all_to_all_tables()
 
     @property
@@ -128,6 +137,36 @@ class ArrowDecompositionMPI:
             r_cnt, r_rows = tables.routing_recv_tables(
                 tn_i, w, own_r, int(self.n_blocks[i + 1]), P)
             self._backward[i + 1] = _Exchange(s_cnt, s_rows, r_cnt, r_rows, eng_r.backend)
+        if P == 1 and L > 1:
+            self._compute_fold_maps()
+
+    def _compute_fold_maps(self) -> None:
+        """Compose the per-pair permutations into direct part-i -> part-0
+        maps (single process only). Forward (features): part-i row s holds
+        X_0[M_i[s]] — from to_next[i-1][r] = s (X_{i}[tn[r]] = X_{i-1}[r],
+        the net effect of _propagate_features at P=1). Backward (results):
+        part-i C row s accumulates into part-0 C row R_i[s] — from
+        to_prev[i][s] (C_{i-1}[tp[s]] += C_i[s], the net effect of
+        _aggregate at P=1). Sentinel/out-of-range entries map to -1."""
+        w = self.width
+        L = self.decomposition_length
+        n0 = int(self.n_blocks[0]) * w
+        M_prev = np.arange(n0, dtype=np.int64)
+        R_prev = np.arange(n0, dtype=np.int64)
+        self._fold_maps = [None]
+        for i in range(1, L):
+            ni = int(self.n_blocks[i]) * w
+            n_prev = int(self.n_blocks[i - 1]) * w
+            tn = np.asarray(self._to_next[i - 1], dtype=np.int64)
+            tp = np.asarray(self._to_prev[i], dtype=np.int64)
+            fmap = np.full(ni, -1, dtype=np.int64)
+            valid = tn < ni
+            fmap[tn[valid]] = np.flatnonzero(valid)
+            M_i = np.where(fmap >= 0, M_prev[np.clip(fmap, 0, None)], -1)
+            dmap = np.where(tp < n_prev, tp, -1)
+            R_i = np.where(dmap >= 0, R_prev[np.clip(dmap, 0, None)], -1)
+            self._fold_maps.append((M_i, R_i))
+            M_prev, R_prev = M_i, R_i
 
     def _checked_slice(self, perm_slice: np.ndarray, engine: ArrowSlimMPI) -> np.ndarray:
         """Validate a per-rank permutation slice against the rank's span."""
@@ -140,19 +179,109 @@ class ArrowDecompositionMPI:
 
     def load_data_from_blocks(self, blocked) -> None:
         """blocked: one block grid per decomposition part
-        (reference arrow_dec_mpi.py:179-181)."""
+        (reference arrow_dec_mpi.py:179-181).
+
+        Folded-permutation mode (ARROW_FOLD, single process, L > 1): parts
+        i >= 1 are re-indexed into part 0's numbering through the composed
+        permutation maps and kept as extra resident structures instead of
+        separate engines + per-step exchanges (see _build_folded). Falls
+        back to the sequential path when the permutation chain does not
+        cover every referenced row (the exchange would read rows no sender
+        wrote)."""
         assert len(blocked) == self.decomposition_length
+        import os
+        fold_env = os.environ.get('ARROW_FOLD', 'auto')
+        want_fold = (self._fold_maps is not None and self.comm.size == 1
+                     and len(blocked) > 1
+                     and (fold_env == '1'
+                          or (fold_env == 'auto' and self.device == 'gpu')))
+        if want_fold:
+            self.engines[0].load_sparse_matrix_from_blocks(blocked[0])
+            folded, dropped = self._build_folded(blocked)
+            if folded is not None:
+                self._folded = folded
+                return
+            import warnings
+            warnings.warn(
+                f"ARROW_FOLD: {dropped} entries reference rows outside the "
+                f"composed permutation chain; falling back to the "
+                f"sequential exchange path")
+            for eng, blocks in zip(self.engines[1:], blocked[1:]):
+                eng.load_sparse_matrix_from_blocks(blocks)
+            return
         for eng, blocks in zip(self.engines, blocked):
             eng.load_sparse_matrix_from_blocks(blocks)
 
+    def _build_folded(self, blocked):
+        """Build one re-indexed structure per part i >= 1: entry
+        (r, c, v) of part i becomes (R_i[r], M_i[c], v) — a direct
+        contribution C_0[R_i[r]] += v * X_0[M_i[c]] — so part i runs as one
+        beta=1 SpMM against part 0's buffers with NO permutation exchange.
+        Per output row the accumulation order equals the sequential path's
+        (each part-i row is summed as its own work item, then added once).
+        Returns (list, 0) or (None, n_dropped) when entries fall outside
+        the composed maps."""
+        w = self.width
+        eng0 = self.engines[0]
+        be = eng0.backend
+        n0 = int(self.n_blocks[0]) * w
+        gpu = be.device == 'cuda'
+        folded = []
+        for i in range(1, self.decomposition_length):
+            M_i, R_i = self._fold_maps[i]
+            rows_l, cols_l, data_l = [], [], []
+            for br, rowlist in enumerate(blocked[i]):
+                for bc, blk in enumerate(rowlist):
+                    if blk is None:
+                        continue
+                    csr = blk.tocsr()
+                    rows_l.append(np.repeat(
+                        np.arange(csr.shape[0], dtype=np.int64),
+                        np.diff(csr.indptr)) + br * w)
+                    cols_l.append(csr.indices.astype(np.int64) + bc * w)
+                    data_l.append(csr.data.astype(np.float32))
+            rows = np.concatenate(rows_l) if rows_l else np.empty(0, np.int64)
+            cols = np.concatenate(cols_l) if cols_l else np.empty(0, np.int64)
+            data = np.concatenate(data_l) if data_l else np.empty(0, np.float32)
+            nr = R_i[np.clip(rows, 0, None)]
+            nc = M_i[np.clip(cols, 0, None)]
+            keep = (nr >= 0) & (nc >= 0)
+            n_drop = int(rows.size - keep.sum())
+            if n_drop:
+                return None, n_drop
+            # group by target row; explicit row_ids keep writes exclusive
+            order = np.argsort(nr, kind='stable')
+            nr, nc, data = nr[order], nc[order], data[order]
+            uniq, counts = np.unique(nr, return_counts=True)
+            indptr = np.zeros(uniq.size + 1, dtype=np.int64)
+            np.cumsum(counts, out=indptr[1:])
+            if gpu:
+                h = be.upload_arrays((uniq.size, n0), indptr,
+                                     nc.astype(np.int32), data,
+                                     row_ids=uniq.astype(np.int64))
+                h.x_rows = n0
+                folded.append(h)
+            else:
+                from scipy import sparse
+                csr = sparse.csr_matrix(
+                    (data.astype(be.np_dtype), nc.astype(np.int64), indptr),
+                    shape=(uniq.size, n0))
+                folded.append((csr, uniq))
+        return folded, 0
+
     def zero_rhs(self, width: int, n_features: int, dtype=np.float32) -> None:
-        for eng in self.engines:
+        for i, eng in enumerate(self.engines):
+            if self._folded is not None and i > 0:
+                continue  # folded parts have no buffers of their own
             eng.zero_rhs(width, n_features, dtype=dtype)
 
     # -- iteration -----------------------------------------------------------
 
     def step(self) -> None:
         """One X <- A @ X iteration (reference arrow_dec_mpi.py:283-307)."""
+        if self._folded is not None:
+            self._step_folded()
+            return
         tic = time.perf_counter()
         self._propagate_features()
         wb_logging.log({"spmm_bcast_time": time.perf_counter() - tic})
@@ -165,6 +294,29 @@ class ArrowDecompositionMPI:
         tic = time.perf_counter()
         self._aggregate()
         wb_logging.log({"spmm_reduce_time": time.perf_counter() - tic})
+
+    def _step_folded(self) -> None:
+        """Folded iteration: part 0's arrow SpMM, then each folded part as
+        one beta=1 SpMM reading part 0's pre-step X and accumulating into
+        part 0's fresh C — results identical to the sequential
+        propagate/spmm/aggregate cascade (same per-row sums; fp32 add order
+        across parts preserved: part i is added after part i-1's own
+        contribution, as the backward cascade does)."""
+        eng0 = self.engines[0]
+        X_pre = eng0.feature_tile()
+        tic = time.perf_counter()
+        eng0.spmm()
+        for h in self._folded:
+            if eng0.backend.device == 'cuda':
+                eng0._timed(lambda h=h: eng0.backend.spmm_block(
+                    h, X_pre, eng0.C_i, 1), h.nnz, h.shape[0], h.x_rows)
+            else:
+                csr, rid = h
+                eng0.C_i.numpy()[rid] += csr @ X_pre.numpy()
+        # C was mutated after the C_0 head was captured: any cached X_0
+        # (the allreduce_x0 fast path) is stale now
+        eng0._x0_valid = False
+        wb_logging.log({'spmm_arrow_time': time.perf_counter() - tic})
 
     def _propagate_features(self) -> None:
         """Forward: route matrix i's features to matrix i+1 along the

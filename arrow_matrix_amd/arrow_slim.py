@@ -197,12 +197,14 @@ class ArrowSlimMPI(ArrowMatrix):
         self._rest_row_offset = 0
         if nw == 0:
             return
-        # opt-in: measured a wash at 20M rows and -8% at 100M vs the
-        # two-launch layout (profiles/r01_*) — the X-stripe re-read the
-        # fusion saves is apparently absorbed by L3/scheduling overlap,
-        # while the merged structure loses locality
+        # DEFAULT ON at world=1 since round 2: under the per-XCD queue
+        # scheduler the single fused launch reads the X stripe ONCE per
+        # iteration (the round-1 "-8%" verdict predates the queue
+        # scheduler): measured +12% at 100M (4435 vs 3966 GF/s) and +22%
+        # at 20M (gpurun_out/r02_ab.log). ARROW_FUSE_ALL=0 restores the
+        # two-launch layout (required by ARROW_SPLIT_COL/ARROW_PAR_ROW0).
         import os as _os
-        fuse_all = (_os.environ.get('ARROW_FUSE_ALL', '0') == '1'
+        fuse_all = (_os.environ.get('ARROW_FUSE_ALL', '1') != '0'
                     and self.comm.size == 1 and self.first_block == 0
                     and self.n_owned == self.tiles_per_side)
         # --- row-0 merge: C_0 = [A_0,first .. A_0,last-1] @ X_stripe -------

@@ -428,7 +428,7 @@ def main():
         # read it, and C is charged once per row written. The instrumented
         # pass covers `instr_steps` iterations.
         instr_steps = 2
-        stripe_rows = sum(eng.n_owned * eng.width for eng in arrow.engines)
+        stripe_rows = sum(eng.n_owned for eng in arrow.engines) * w
         xc_bytes = instr_steps * 4.0 * k * (2 * stripe_rows)  # X once + C once
         total_bytes = a_meta_bytes + xc_bytes
         achieved = total_bytes / (total_ms / 1e3) / 1e9  # GB/s

@@ -136,3 +136,50 @@ def test_petsc_cpu_float64():
     Y = eng.spmm(X)
     assert Y.dtype == torch.float64
     np.testing.assert_allclose(Y.numpy(), A @ X, rtol=1e-12, atol=1e-12)
+
+
+def test_engine_folded_permutation_cpu(monkeypatch):
+    """Folded-permutation mode (ARROW_FOLD=1, single process, L>1): parts
+    i>=1 re-indexed into part 0's numbering, no per-step exchange — results
+    must match the sequential cascade / golden (arrow_dec.py::_build_folded)."""
+    monkeypatch.setenv('ARROW_FOLD', '1')
+    for n_blocks, seed in ([4, 2], 3), ([3, 3], 4), ([4, 3, 2], 5):
+        decomp = synth.synth_arrow_decomposition(5, n_blocks, avg_deg=5, seed=seed)
+        results, goldens = _run_engine(decomp, 5, n_blocks, 8, iters=3, seed=seed)
+        for C, G in zip(results, goldens):
+            np.testing.assert_allclose(C, G, rtol=1e-4, atol=1e-4)
+
+
+def test_engine_folded_engages_and_matches_sequential(monkeypatch):
+    """The fold path must actually engage (structures built, exchange
+    skipped) and agree with the sequential path on the same input."""
+    import tempfile
+    width, n_blocks, k = 6, [4, 2], 5
+    decomp = synth.synth_arrow_decomposition(width, n_blocks, avg_deg=5, seed=21)
+    n = n_blocks[0] * width
+    rng = np.random.default_rng(21)
+    X = (2 * rng.random((n, k)) - 1).astype(np.float32)
+
+    def run(fold):
+        monkeypatch.setenv('ARROW_FOLD', '1' if fold else '0')
+        with tempfile.TemporaryDirectory() as td:
+            prefix = os.path.join(td, 'g')
+            graphio.save_decomposition_new(decomp, prefix, width)
+            blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+                None, prefix, width)
+            arrow = ArrowDecompositionMPI.initialize(None, nb, tp, tn, width,
+                                                     k, device='cpu')
+            arrow.load_data_from_blocks(blocks)
+            arrow.zero_rhs(width, k)
+            assert (arrow._folded is not None) == fold
+            arrow.B.set_features(X[decomp[0][1]].copy())
+            outs = []
+            for _ in range(2):
+                arrow.step()
+                outs.append(arrow.B.allgather_result().copy())
+                arrow.B.set_features(arrow.B.result_tile())
+            return outs
+
+    folded, seq = run(True), run(False)
+    for F, S in zip(folded, seq):
+        np.testing.assert_allclose(F, S, rtol=2e-6, atol=2e-6)

@@ -200,3 +200,61 @@ def test_engine_gpu_rest_chunked(monkeypatch):
                                    device='gpu', seed=29)
     for C, G in zip(results, goldens):
         np.testing.assert_allclose(C, G, rtol=2e-4, atol=2e-4)
+
+
+@pytest.mark.gpu
+def test_engine_gpu_multipart_sequential_exchange(monkeypatch):
+    """ARROW_FOLD=0: the sequential permutation-exchange path (device
+    permute/scatter kernels) must stay correct at L>1 on GPU now that the
+    folded path is the single-process default."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    monkeypatch.setenv('ARROW_FOLD', '0')
+    from tests.test_engine_cpu import _run_engine
+    from arrow_matrix_amd import synth
+    decomp = synth.synth_arrow_decomposition(50, [4, 2], avg_deg=8, seed=2)
+    results, goldens = _run_engine(decomp, 50, [4, 2], 8, iters=2,
+                                   device='gpu', seed=2)
+    for C, G in zip(results, goldens):
+        np.testing.assert_allclose(C, G, rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.gpu
+def test_engine_gpu_folded_matches_sequential(monkeypatch):
+    """GPU fold parity: folded vs sequential exchange on the same L=3
+    decomposition."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import os
+    import tempfile
+    from arrow_matrix_amd import graphio, synth
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+    width, n_blocks, k = 40, [4, 3, 2], 16
+    decomp = synth.synth_arrow_decomposition(width, n_blocks, avg_deg=6, seed=33)
+    n = n_blocks[0] * width
+    rng = np.random.default_rng(33)
+    X = (2 * rng.random((n, k)) - 1).astype(np.float32)
+
+    def run(fold):
+        monkeypatch.setenv('ARROW_FOLD', '1' if fold else '0')
+        with tempfile.TemporaryDirectory() as td:
+            prefix = os.path.join(td, 'g')
+            graphio.save_decomposition_new(decomp, prefix, width)
+            blocks, nb, tp, tn = ArrowDecompositionMPI.load_decomposition_new(
+                None, prefix, width)
+            arrow = ArrowDecompositionMPI.initialize(None, nb, tp, tn, width,
+                                                     k, device='gpu')
+            arrow.load_data_from_blocks(blocks)
+            arrow.zero_rhs(width, k)
+            assert (arrow._folded is not None) == fold
+            arrow.B.set_features(X[decomp[0][1]].copy())
+            outs = []
+            for _ in range(2):
+                arrow.step()
+                outs.append(arrow.B.allgather_result().copy())
+                arrow.B.set_features(arrow.B.result_tile())
+            return outs
+
+    folded, seq = run(True), run(False)
+    for F, S in zip(folded, seq):
+        np.testing.assert_allclose(F, S, rtol=1e-5, atol=1e-5)
