@@ -278,22 +278,60 @@ class ArrowDecompositionMPI:
     # -- iteration -----------------------------------------------------------
 
     def step(self) -> None:
-        """One X <- A @ X iteration (reference arrow_dec_mpi.py:283-307)."""
+        """One X <- A @ X iteration (reference arrow_dec_mpi.py:283-307).
+        At world>1 with L>1 the forward exchange for pair (i, i+1) is
+        POSTED asynchronously and part i's SpMM runs while the transfer is
+        in flight — the reference's own overlap (the Ialltoallv posted at
+        arrow_dec_mpi.py:304 is waited only after the local compute at
+        :306). Data dependencies (and results) are identical to the
+        sequential schedule."""
         if self._folded is not None:
             self._step_folded()
             return
-        tic = time.perf_counter()
-        self._propagate_features()
-        wb_logging.log({"spmm_bcast_time": time.perf_counter() - tic})
-
-        tic = time.perf_counter()
-        for eng in self.engines:
-            eng.spmm()
-        wb_logging.log({'spmm_arrow_time': time.perf_counter() - tic})
+        if self.comm.size > 1 and self.decomposition_length > 1:
+            tic = time.perf_counter()
+            self._step_overlapped_forward()
+            wb_logging.log({'spmm_arrow_time': time.perf_counter() - tic})
+        else:
+            tic = time.perf_counter()
+            self._propagate_features()
+            wb_logging.log({"spmm_bcast_time": time.perf_counter() - tic})
+            tic = time.perf_counter()
+            for eng in self.engines:
+                eng.spmm()
+            wb_logging.log({'spmm_arrow_time': time.perf_counter() - tic})
 
         tic = time.perf_counter()
         self._aggregate()
         wb_logging.log({"spmm_reduce_time": time.perf_counter() - tic})
+
+    def _step_overlapped_forward(self) -> None:
+        """Forward exchange posted async, overlapped with the previous
+        part's SpMM (world>1). Same dependency chain as
+        _propagate_features + the spmm loop: exchange i reads part i's
+        features (set by exchange i-1), part i+1's features are set before
+        its spmm."""
+        L = self.decomposition_length
+        pending = None
+        for i in range(L):
+            if pending is not None:
+                ex, recvbuf, works = pending
+                for wk in works:
+                    wk.wait()
+                eng_r = self.engines[i]
+                eng_r.backend.scatter_rows(eng_r.C_i, ex.recv_rows, recvbuf)
+                eng_r._x0_valid = False
+                eng_r.set_features(eng_r.C_i)
+                pending = None
+            if i < L - 1:
+                ex = self._forward[i]
+                eng_s = self.engines[i]
+                sendbuf = eng_s.backend.gather_rows(eng_s.feature_tile(),
+                                                    ex.send_rows)
+                recvbuf, works = self.comm.alltoallv_async(
+                    sendbuf, ex.send_counts, ex.recv_counts)
+                pending = (ex, recvbuf, works)
+            self.engines[i].spmm()
 
     def _step_folded(self) -> None:
         """Folded iteration: part 0's arrow SpMM, then each folded part as

@@ -92,10 +92,12 @@ def _worker(rank, port, tmpdir, result_q, world=2):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2, 3])
+@pytest.mark.parametrize("world", [2, 3, 8])
 def test_distributed_gloo(world):
     # world=3 exercises uneven spans including a rank owning NO blocks of
-    # small parts (nb=2 over 3 ranks)
+    # small parts (nb=2 over 3 ranks); world=8 is the full-node layout the
+    # driver's 8-GPU scale bench runs (every case has ranks owning no
+    # blocks of some parts — the collective-schedule parity path)
     port = _free_port()
     ctx = mp.get_context('spawn')
     q = ctx.Queue()
