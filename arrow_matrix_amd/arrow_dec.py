@@ -191,10 +191,18 @@ class ArrowDecompositionMPI:
         assert len(blocked) == self.decomposition_length
         import os
         fold_env = os.environ.get('ARROW_FOLD', 'auto')
+        # auto: fold only in the LAUNCH-BOUND small regime (part-0 rows <=
+        # 4M). At larger scale the sequential path wins: materialising the
+        # permuted X deduplicates the ~deg-many reads each part-i entry
+        # would otherwise do through a random permutation (measured at 20M
+        # rows L=2: fold 3060 vs sequential 3155 GF/s, and the gap grows
+        # with size — profiles/r02_ab2_sweep.log L2_fold/L2_seq).
+        n0_rows = int(self.n_blocks[0]) * self.width
         want_fold = (self._fold_maps is not None and self.comm.size == 1
                      and len(blocked) > 1
                      and (fold_env == '1'
-                          or (fold_env == 'auto' and self.device == 'gpu')))
+                          or (fold_env == 'auto' and self.device == 'gpu'
+                              and n0_rows <= 4_000_000)))
         if want_fold:
             self.engines[0].load_sparse_matrix_from_blocks(blocked[0])
             folded, dropped = self._build_folded(blocked)

@@ -633,29 +633,34 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
     return (e && e[0] == '0') ? 0 : 1;
   }();
   const int qd = blk.queue_mode >= 0 ? blk.queue_mode : env_queue_default();
-  // ARROW_QWAVE=1: per-wave queue grabs (spmm_kernel_qw) — no per-chunk
-  // block barrier, so the queue scheduler also pays at GROUP=4 (k=16)
-  static const int qwave = [] {
+  // Per-wave queue grabs (spmm_kernel_qw): no per-chunk block barrier.
+  // Default AUTO: wave variant at GROUP < 8 (k=16: measured +10% over the
+  // fused grid-stride and +15% over the round-1 baseline,
+  // profiles/r02_ab2_sweep.log k16_fused_q1_qw), block variant at
+  // GROUP >= 8 (k=128: wave measured -17%, r02_ab1). ARROW_QWAVE=0/1
+  // forces.
+  static const int qwave_env = [] {
     const char *e = getenv("ARROW_QWAVE");
-    return (e && e[0] == '1') ? 1 : 0;
+    if (!e || !e[0]) return -1;
+    return (e[0] == '1') ? 1 : 0;
   }();
-  // policy default: queues need wide groups AND enough work per structure —
+  const bool qwave = qwave_env >= 0 ? (qwave_env == 1) : (GROUP < 8);
+  // policy default: queues need GROUP >= 4 AND enough work per structure —
   // below ~32M nnz (ARROW_Q_MIN_NNZ overrides) the per-grab overhead
-  // outweighs the L2 window benefit; the wave variant extends the policy
-  // to GROUP >= 4
+  // outweighs the L2 window benefit
   static const int64_t q_min_nnz = [] {
     const char *e = getenv("ARROW_Q_MIN_NNZ");
     return e ? (int64_t)atoll(e) : (32LL << 20);
   }();
   const bool useq = blk.qseg &&
-      (qd >= 0 ? qd : ((GROUP >= 8 || (qwave && GROUP >= 4)) &&
-                       blk.nnz >= q_min_nnz));
+      (qd >= 0 ? qd : (GROUP >= 4 && blk.nnz >= q_min_nnz));
   // Queue mode: size the grid to residency (8 blocks/CU fit at this
   // occupancy — 4 waves/WG, 8 waves/SIMD), not to the item count; chunk =
-  // 2 rounds per grab keeps each XCD's in-flight row window a few MB.
+  // 4 rounds per grab (measured best on the fused cfg4 structure: 1/2/4 →
+  // 4306/4450/4552 GF/s, profiles/r02_ab2_sweep.log).
   static const int q_chunk_mult = [] {
     const char *e = getenv("ARROW_Q_CHUNK");
-    return e ? std::max(1, atoi(e)) : 2;
+    return e ? std::max(1, atoi(e)) : 4;
   }();
   static const int q_blocks_env = [] {
     const char *e = getenv("ARROW_Q_BLOCKS");

@@ -169,7 +169,11 @@ def _kernel_name(k, nnz_dominant):
     while group < lanes and group < 64:
         group <<= 1
     big = nnz_dominant >= (32 << 20)
-    return "spmm_kernel_q" if (group >= 8 and big) else "spmm_kernel"
+    if group >= 8 and big:
+        return "spmm_kernel_q"
+    if group >= 4 and big:
+        return "spmm_kernel_qw"  # per-wave grabs at small GROUP (k=16)
+    return "spmm_kernel"
 
 
 def _cpu_worker(args):

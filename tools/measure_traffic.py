@@ -70,7 +70,9 @@ def main():
     a = ap.parse_args()
     bench_args = ['--rows', a.rows, '--features', a.features, '--band', a.band,
                   '--steps', a.steps, '--warmup', a.warmup]
-    base = os.path.join(REPO, 'gpurun_out')
+    # rocprof DBs are large — keep them OUT of gpurun_out (its merge-back
+    # is capped at 64 MiB); only the small summary json goes there
+    base = os.path.join('/tmp', 'arrow_traffic')
     fetch = run_pass(['FETCH_SIZE'], os.path.join(base, 'tr_fetch'), bench_args)
     write = run_pass(['WRITE_SIZE'], os.path.join(base, 'tr_write'), bench_args)
     fvals = sorted(fetch.get('FETCH_SIZE', {}).items())
