@@ -19,11 +19,14 @@
 //     item, so beta=0 also zeroes empty rows.
 //   * A's (col, val) stream is read once per item by all lanes of the
 //     group (same-address broadcast within the wave's transaction).
-//   * Work items are consumed either by a grid-stride loop (small GROUP)
-//     or by the per-XCD queue scheduler (GROUP >= 8 by default): 8
-//     contiguous nnz-balanced item segments drained via per-XCD atomic
-//     chunk counters so each XCD walks one tight X window in order
-//     (see spmm_kernel_q below; measured +54 % at k=128).
+//   * Work items are consumed either by a grid-stride loop (small
+//     structures) or by the per-XCD queue scheduler (GROUP >= 4 and
+//     >= 32M nnz by default): 8 contiguous nnz-balanced item segments
+//     drained via per-XCD atomic chunk counters so each XCD walks one
+//     tight X window in order (measured +54 % at k=128). Grabs are per
+//     WORKGROUP at GROUP >= 8 (spmm_kernel_q) and per WAVE at GROUP < 8
+//     (spmm_kernel_qw — no __syncthreads rendezvous; k=16 measured
+//     +15 % over grid-stride, profiles/r02_ab2_sweep.log).
 
 #include <hip/hip_runtime.h>
 #include <algorithm>
