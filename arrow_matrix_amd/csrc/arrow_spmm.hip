@@ -671,9 +671,15 @@ int launch_spmm_vg(const CsrBlock &blk, const float *X0, const float *X1,
   }();
   const int q_blocks = blk.q_blocks > 0 ? blk.q_blocks : q_blocks_env;
   const int chunk_items = GROUPS_PER_BLOCK * q_chunk_mult;
-  // per-wave grabs pull GROUPS_PER_WAVE-item rounds; same default round
-  // count per grab as the block variant
-  const int chunk_items_w = (64 / GROUP) * q_chunk_mult;
+  // per-wave grabs pull GROUPS_PER_WAVE-item rounds. Default 2 rounds —
+  // the wave variant regresses at larger grabs (k=16: 1811 GF/s at 2 vs
+  // 1342 at 4, 786 at 8 — its locality window is per-wave, so bigger
+  // chunks smear it). ARROW_QW_CHUNK overrides.
+  static const int qw_chunk_mult = [] {
+    const char *e = getenv("ARROW_QW_CHUNK");
+    return e ? std::max(1, atoi(e)) : 2;
+  }();
+  const int chunk_items_w = (64 / GROUP) * qw_chunk_mult;
   int blocks = (int)std::min<int64_t>(
       (blk.n_items + GROUPS_PER_BLOCK - 1) / GROUPS_PER_BLOCK,
       useq ? q_blocks : 8192);
