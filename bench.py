@@ -63,7 +63,13 @@ def build_parser():
     p.add_argument('--json-out', type=str, default=None)
     p.add_argument('--graph', choices=['auto', 'on', 'off'], default='auto',
                    help='hipGraph-capture the steady-state iteration '
-                        '(BASELINE cfg5); auto = on for single-GPU runs')
+                        '(BASELINE cfg5); auto = on for single-GPU runs. '
+                        'world>1 stays capture-less by default: capturing '
+                        'RCCL collectives could not be validated on a '
+                        '1-GPU rig (RCCL 2.26 refuses two ranks per '
+                        'device, profiles/r02_ab3_summary.log) and the '
+                        'chunk-pipelined C_0 allreduce already overlaps '
+                        'the collective with compute')
     return p
 
 
