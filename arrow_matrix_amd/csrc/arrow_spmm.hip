@@ -616,6 +616,11 @@ LaunchCfg pick_cfg(int64_t k) {
       [] { const char *e = getenv("ARROW_SPMM_G64");
            return (e && e[0] == '1') ? 1 : 0; }();
   if (g64 && k == 128) return {2, 64, 128};  // A/B: wave-wide groups, float2
+  static const int k16g8 =
+      [] { const char *e = getenv("ARROW_K16_G8");
+           return (e && e[0] == '1') ? 1 : 0; }();
+  if (k16g8 && k == 16) return {2, 8, 16};  // A/B: float2 x 8 lanes,
+                                            // block-grab queue eligible
   int vec = (k % 4 == 0) ? 4 : (k % 2 == 0) ? 2 : 1;
   int64_t lanes_needed = (k + vec - 1) / vec;
   int group = 1;
