@@ -657,6 +657,11 @@ class ArrowSlimMPI(ArrowMatrix):
         if self.comm.size == 1 or self.n_owned == 0:
             return
         import torch.distributed as dist
+        if (self.X_i.is_cuda
+                and dist.get_backend(getattr(self.comm, 'group', None)) == 'gloo'):
+            # gloo has no CUDA p2p: stage halos through host (validation
+            # rigs only; the product multi-GPU path is RCCL)
+            return self._exchange_halos_staged()
         w = self.width
         P = self.comm.size
         bpr = self.blocks_per_rank
@@ -678,6 +683,36 @@ class ArrowSlimMPI(ArrowMatrix):
         if ops:
             for wk in dist.batch_isend_irecv(ops):
                 wk.wait()
+
+    def _exchange_halos_staged(self) -> None:
+        """Halo exchange with host staging (gloo + CUDA tensors)."""
+        import torch.distributed as dist
+        w = self.width
+        P = self.comm.size
+        bpr = self.blocks_per_rank
+        prev_rank = (self.first_block - 1) // bpr if self.first_block >= 1 else -1
+        next_owner = self.last_block // bpr if self.last_block < self.tiles_per_side else -1
+        group = getattr(self.comm, 'group', None)
+        ops = []
+        lo_cpu = self.X_halo_lo.cpu() if prev_rank >= 0 else None
+        hi_cpu = self.X_halo_hi.cpu() if (0 <= next_owner < P) else None
+        if prev_rank >= 0:
+            ops.append(dist.P2POp(dist.isend, self.X_i[:w].cpu(), prev_rank,
+                                  group=group))
+            ops.append(dist.P2POp(dist.irecv, lo_cpu, prev_rank, group=group))
+        if 0 <= next_owner < P:
+            ops.append(dist.P2POp(
+                dist.isend,
+                self.X_i[(self.n_owned - 1) * w:self.n_owned * w].cpu(),
+                next_owner, group=group))
+            ops.append(dist.P2POp(dist.irecv, hi_cpu, next_owner, group=group))
+        if ops:
+            for wk in dist.batch_isend_irecv(ops):
+                wk.wait()
+        if lo_cpu is not None:
+            self.X_halo_lo.copy_(lo_cpu)
+        if hi_cpu is not None:
+            self.X_halo_hi.copy_(hi_cpu)
 
     def _spmm_cpu(self) -> None:
         """Per-block scipy dataflow — the reference's own cpu path

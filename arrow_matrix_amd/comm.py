@@ -115,6 +115,11 @@ class TorchDistComm(Comm):
         if send.device.type == 'cpu' and dist.get_backend(self.group) == 'nccl':
             recv = self.alltoallv(send.cuda(), send_counts, recv_counts)
             return recv.cpu()
+        if send.device.type == 'cuda' and dist.get_backend(self.group) == 'gloo':
+            # gloo has no CUDA p2p: stage through host (validation rigs
+            # only; the product multi-GPU path is RCCL)
+            recv = self.alltoallv(send.cpu(), send_counts, recv_counts)
+            return recv.to(send.device)
         sdispl = [0]
         for c in send_counts:
             sdispl.append(sdispl[-1] + int(c))
@@ -132,6 +137,8 @@ class TorchDistComm(Comm):
         """Post the exchange and return (recv, works) WITHOUT waiting —
         kernels enqueued before the waits overlap with the transfer
         (RCCL runs on its own stream; wait() inserts stream deps)."""
+        if send.device.type == 'cuda' and dist.get_backend(self.group) == 'gloo':
+            return self.alltoallv(send, send_counts, recv_counts), []
         sdispl = [0]
         for c in send_counts:
             sdispl.append(sdispl[-1] + int(c))
