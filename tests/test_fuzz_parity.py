@@ -103,3 +103,11 @@ def test_fuzz_case_gpu(case):
                        for B, p in decomp]
         golden = compute_spmm(zero_decomp, X)[perm0]
         np.testing.assert_allclose(C, golden, rtol=2e-4, atol=2e-4)
+
+
+@pytest.mark.parametrize("case", [c for c in CASES if len(c['n_blocks']) > 1])
+def test_fuzz_case_folded(case, monkeypatch):
+    """The same randomised L>1 sweep through the FOLDED permutation path
+    (arrow_dec._build_folded) on CPU."""
+    monkeypatch.setenv('ARROW_FOLD', '1')
+    test_fuzz_case(case)
