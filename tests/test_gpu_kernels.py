@@ -295,3 +295,60 @@ def test_spmm_col_items_row_ids(gpu):
         for t in range(indptr[r], indptr[r + 1]):
             ref[row_ids[r]] += float(vals[t]) * X[idx[t]]
     np.testing.assert_allclose(Ct.cpu().numpy(), ref, rtol=2e-5, atol=2e-5)
+
+
+@pytest.mark.parametrize("k", [4, 8, 16, 32, 128])
+def test_spmm_queue_wave_grabs(gpu, k, monkeypatch):
+    """Per-WAVE queue grabs (spmm_kernel_qw, ARROW_QWAVE=1): identical
+    results across k (the round-2 default scheduler for GROUP < 8)."""
+    monkeypatch.setenv('ARROW_QWAVE', '1')
+    A = _random_csr(600, 800, 0.02, seed=300 + k)
+    _check_spmm_queue(gpu, A, k, beta=0, seed=40 + k)
+    _check_spmm_queue(gpu, A, k, beta=1, seed=41 + k)
+
+
+def test_spmm_queue_wave_grabs_hub(gpu, monkeypatch):
+    """Wave grabs with split (atomic) hub rows and empty rows."""
+    monkeypatch.setenv('ARROW_QWAVE', '1')
+    rows, cols, k = 400, 5000, 16
+    rs = np.random.RandomState(13)
+    A = sparse.random(rows, cols, density=0.002, format='lil', random_state=rs,
+                      dtype=np.float64)
+    A[3, :] = rs.rand(cols)
+    A[111, :] = 0
+    A = sparse.csr_matrix(A, dtype=np.float32)
+    _check_spmm_queue(gpu, A, k, beta=0, seed=13, rtol=2e-5, atol=2e-5)
+
+
+def test_spmm_k16_g8_layout(gpu, monkeypatch):
+    """ARROW_K16_G8=1 (float2 x 8-lane groups at k=16): identical results
+    with and without the queue scheduler."""
+    monkeypatch.setenv('ARROW_K16_G8', '1')
+    A = _random_csr(500, 700, 0.03, seed=77)
+    _check_spmm_queue(gpu, A, 16, beta=0, seed=77)
+    _check_spmm_queue(gpu, A, 16, beta=1, seed=78)
+
+
+def test_spmm_within_segment_col_sort(gpu):
+    """Two-level item order (arrow_csr_create_opts flags=2): identical
+    results on a hub structure (row segments kept, columns sorted inside)."""
+    from arrow_matrix_amd import hip
+    rows, cols, k = 300, 4000, 32
+    rs = np.random.RandomState(17)
+    A = sparse.random(rows, cols, density=0.01, format='lil', random_state=rs,
+                      dtype=np.float64)
+    A[5, :] = rs.rand(cols)
+    A = sparse.csr_matrix(A, dtype=np.float32)
+    rng = np.random.default_rng(17)
+    X = (2 * rng.random((cols, k)) - 1).astype(np.float32)
+    Xt = torch.from_numpy(X).cuda()
+    blk = hip.CsrBlockGPU(arrays=((rows, cols), A.indptr.astype(np.int64),
+                                  A.indices.astype(np.int32), A.data),
+                          col_items=2)
+    blk.set_queue(1)
+    Ct = torch.zeros((rows, k), device='cuda')
+    blk.spmm(Xt.data_ptr(), Ct.data_ptr(), k, 0,
+             torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    ref = A @ X
+    np.testing.assert_allclose(Ct.cpu().numpy(), ref, rtol=2e-5, atol=2e-5)
