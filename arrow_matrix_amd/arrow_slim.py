@@ -83,6 +83,13 @@ class ArrowSlimMPI(ArrowMatrix):
         self.allreduce_x0 = False
         self._x0_valid = False
         self._prev_result = None
+        # deferred C_0-allreduce pipeline (allreduce_x0 + world>1, slim):
+        # the collective posted in step t is waited in step t+1 just before
+        # its first consumer, so it overlaps BOTH the rest launch of step t
+        # and the row-0 compute of step t+1 (ranks owning block 0 wait
+        # first and refresh their stripe head from the reduced X_0 — the
+        # only rows whose consumer is the row-0 launch itself)
+        self._pending_x0_works: list = []
         # merged resident GPU structures (built at load time, gpu only)
         self._A_row0 = None
         self._A_rest = None
@@ -516,6 +523,13 @@ class ArrowSlimMPI(ArrowMatrix):
         bcast_work = None
         x0_ready = (self.allreduce_x0 and self._x0_valid
                     and self.X_i is self._prev_result)
+        if self._pending_x0_works and not x0_ready:
+            # features were replaced outside the iterated loop while a
+            # deferred allreduce is still in flight on X_0 — drain it
+            # before the buffer is overwritten below
+            for wk in self._pending_x0_works:
+                wk.wait()
+            self._pending_x0_works = []
         if self._A_all is None and not x0_ready:
             if self.first_block == 0 and self.n_owned > 0:
                 self.X_0.copy_(self.X_i[:w])
@@ -535,9 +549,16 @@ class ArrowSlimMPI(ArrowMatrix):
         wb_logging.log({"spmm_kernel_time": time.perf_counter() - tic})
 
         # the reduced C_0 is block-row 0's result (arrow_slim_mpi.py:152-155);
-        # in the fully-fused single-process layout it was written in place
+        # in the fully-fused single-process layout it was written in place.
+        # In the DEFERRED allreduce pipeline the head copy moves to the
+        # start of the next step (_spmm_gpu refreshes the stripe head from
+        # the reduced X_0 before the row-0 launch — the head's only
+        # consumer), so the iterated loop's values are unchanged while the
+        # collective overlaps this step's rest launch AND the next step's
+        # row-0 compute.
         if (self._A_all is None and self.first_block == 0
-                and self.n_owned > 0 and self.comm.rank == 0):
+                and self.n_owned > 0 and self.comm.rank == 0
+                and not self._pending_x0_works):
             self.C_i[:w].copy_(self.C_0)
         if self.allreduce_x0 and self._A_all is None \
                 and self.backend.device == 'cuda':
@@ -565,6 +586,21 @@ class ArrowSlimMPI(ArrowMatrix):
         inserts stream dependencies, not host blocks)."""
         be = self.backend
         w = self.width
+
+        # deferred-allreduce pipeline: drain LAST step's C_0 collective at
+        # its first consumer. Ranks owning block 0 consume the head rows in
+        # their row-0 launch — wait now and refresh the stripe head from
+        # the reduced X_0 (the deferred equivalent of last step's
+        # C_i[:w].copy_(C_0)); other ranks wait later, just before the
+        # rest launch (their only X_0 consumer) so the collective also
+        # overlaps this step's row-0 compute.
+        pend = self._pending_x0_works
+        if pend and self.first_block == 0 and self.n_owned > 0:
+            for wk in pend:
+                wk.wait()
+            self.X_i[:w].copy_(self.X_0)
+            pend = []
+            self._pending_x0_works = []
 
         if self._A_all is not None:
             # single process: whole matrix in ONE fused launch, C in place;
@@ -636,6 +672,12 @@ class ArrowSlimMPI(ArrowMatrix):
         # C_rest = A_diag_merged @ X_stripe + A_col_merged @ X_0 fused:
         # C written once (vs the reference's multiply-then-accumulate,
         # arrow_slim_mpi.py:121-144)
+        if pend:
+            # ranks without block 0: last step's deferred allreduce must
+            # land before the rest launch reads X_0
+            for wk in pend:
+                wk.wait()
+            self._pending_x0_works = []
         if bcast_work is not None:
             bcast_work.wait()  # rest reads X_0
         if self._A_rest:
@@ -663,9 +705,15 @@ class ArrowSlimMPI(ArrowMatrix):
             # join: the main stream's C_0 consumers (head copy / X_0 swap)
             # wait for the side-stream row-0 work
             torch.cuda.current_stream().wait_event(ev_r)
-        for wk in reduce_works:
-            if wk is not None:
-                wk.wait()
+        defer = (self.allreduce_x0 and self.comm.size > 1
+                 and not self.banded)
+        if defer:
+            self._pending_x0_works = [wk for wk in reduce_works
+                                      if wk is not None]
+        else:
+            for wk in reduce_works:
+                if wk is not None:
+                    wk.wait()
 
     def _exchange_halos(self) -> None:
         """±1 halo X exchange for the banded mode: my first owned tile goes

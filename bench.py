@@ -61,6 +61,12 @@ def build_parser():
                    help='diagonal-block band half-width (planar-like locality)')
     p.add_argument('--no-cpu-baseline', action='store_true')
     p.add_argument('--json-out', type=str, default=None)
+    p.add_argument('--check', action='store_true',
+                   help='parity mode (L=1): run 3 iterations of the exact '
+                        'bench pipeline (incl. the allreduce_x0 deferred '
+                        'collective at world>1) and compare against a '
+                        'scipy golden rebuilt from the deterministic '
+                        'generators; prints a check line and exits')
     p.add_argument('--graph', choices=['auto', 'on', 'off'], default='auto',
                    help='hipGraph-capture the steady-state iteration '
                         '(BASELINE cfg5); auto = on for single-GPU runs. '
@@ -275,6 +281,67 @@ def traffic_probe(args):
     return None
 
 
+def parity_check(arrow, comm, w, nb, k, band, use_gpu, world):
+    """3 iterations of the bench pipeline vs a scipy golden rebuilt from
+    the same deterministic block generators (L=1 only). Every rank's
+    feature stripe is seeded by rank with a numpy rng so rank 0 can
+    reproduce the full X."""
+    import torch
+    from scipy import sparse
+    eng0 = arrow.engines[0]
+    first, last = eng0.first_block, eng0.last_block
+    stripe_np = (2 * np.random.default_rng(100 + comm.rank)
+                 .random(((last - first) * w, k)) - 1).astype(np.float32)
+    if stripe_np.shape[0] == 0:
+        stripe_np = np.zeros((w, k), np.float32)
+    X0 = torch.from_numpy(stripe_np.copy())
+    eng0.set_features(X0.cuda() if use_gpu else X0)
+    for _ in range(3):
+        arrow.step()
+        eng0.set_features(eng0.result_tile())
+    C = eng0.allgather_result()
+    status = {"check": "ok"}
+    if comm.rank == 0:
+        gen_device = 'cuda' if use_gpu else 'cpu'
+        bpr = -(-nb // world)
+        stripes = []
+        for r in range(world):
+            f, l = min(r * bpr, nb), min(min(r * bpr, nb) + bpr, nb)
+            if l > f:
+                stripes.append((2 * np.random.default_rng(100 + r)
+                                .random(((l - f) * w, k)) - 1)
+                               .astype(np.float32))
+        X_full = np.concatenate(stripes)
+        rows_l, cols_l, data_l = [], [], []
+        for c in range(nb):
+            ip, ix, dv = generate_block('row0', w, 10 * c + 1, gen_device, band)
+            rows_l.append(np.repeat(np.arange(w), np.diff(ip)))
+            cols_l.append(ix.astype(np.int64) + c * w)
+            data_l.append(dv)
+        for r in range(1, nb):
+            for role, seed, coff in (('diag', 10 * r + 2, r * w),
+                                     ('col', 10 * r + 3, 0)):
+                ip, ix, dv = generate_block(role, w, seed, gen_device, band)
+                rows_l.append(np.repeat(np.arange(w), np.diff(ip)) + r * w)
+                cols_l.append(ix.astype(np.int64) + coff)
+                data_l.append(dv)
+        A = sparse.csr_matrix(
+            (np.concatenate(data_l),
+             (np.concatenate(rows_l), np.concatenate(cols_l))),
+            shape=(nb * w, nb * w))
+        G = X_full
+        for _ in range(3):
+            G = A @ G
+        err = float(np.abs(C - G).max())
+        scale = max(1.0, float(np.abs(G).max()))
+        status = {"check": "ok" if err <= 1e-4 * scale else "FAIL",
+                  "max_abs_err": err, "scale": scale,
+                  "rel_err": err / scale, "world": world, "rows": nb * w,
+                  "k": k}
+        print(json.dumps(status))
+    return status
+
+
 def main():
     args = build_parser().parse_args()
     import torch
@@ -349,6 +416,13 @@ def main():
         # valid at L == 1: with multiple parts the inter-part exchange
         # rewrites C after the spmm, so next X_0 != this C_0.
         eng.allreduce_x0 = (args.device == 'gpu' and L == 1)
+    if args.check:
+        assert L == 1, "--check supports L=1 (the headline config)"
+        parity_check(arrow, comm, w, nb, k, args.band, use_gpu, comm.size)
+        if world > 1:
+            dist.destroy_process_group()
+        return
+
     if use_gpu:
         g = torch.Generator(device='cuda')
         g.manual_seed(42 + comm.rank)
