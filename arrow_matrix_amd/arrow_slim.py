@@ -705,8 +705,10 @@ class ArrowSlimMPI(ArrowMatrix):
             # join: the main stream's C_0 consumers (head copy / X_0 swap)
             # wait for the side-stream row-0 work
             torch.cuda.current_stream().wait_event(ev_r)
+        import os as _os2
         defer = (self.allreduce_x0 and self.comm.size > 1
-                 and not self.banded)
+                 and not self.banded
+                 and _os2.environ.get('ARROW_X0_DEFER', '1') != '0')
         if defer:
             self._pending_x0_works = [wk for wk in reduce_works
                                       if wk is not None]
@@ -823,10 +825,25 @@ class ArrowSlimMPI(ArrowMatrix):
 
     # -- result --------------------------------------------------------------
 
+    def _flush_x0_pipeline(self) -> None:
+        """Drain the deferred C_0 allreduce (allreduce_x0 world>1 iterated
+        loop) and materialise block-row 0's result head — the deferred
+        equivalent of the skipped C_i[:w].copy_(C_0). Called before any
+        result readout."""
+        if not self._pending_x0_works:
+            return
+        for wk in self._pending_x0_works:
+            wk.wait()
+        self._pending_x0_works = []
+        if self.first_block == 0 and self.n_owned > 0:
+            # after the swap X_0 holds the reduced C_0 of the last step
+            self.C_i[:self.width].copy_(self.X_0)
+
     def allgather_result(self, C=None):
         """All-gathers the per-rank result stripes into the full
         (tiles_per_side*width, k) matrix (reference arrow_slim_mpi.py:415-425).
         Returns numpy; fills C in place if given."""
+        self._flush_x0_pipeline()
         w = self.width
         k = self.C_i.shape[1]
         pad_rows = self.blocks_per_rank * w

@@ -299,46 +299,65 @@ def parity_check(arrow, comm, w, nb, k, band, use_gpu, world):
     for _ in range(3):
         arrow.step()
         eng0.set_features(eng0.result_tile())
-    C = eng0.allgather_result()
-    status = {"check": "ok"}
-    if comm.rank == 0:
-        gen_device = 'cuda' if use_gpu else 'cpu'
-        bpr = -(-nb // world)
-        stripes = []
-        for r in range(world):
-            f, l = min(r * bpr, nb), min(min(r * bpr, nb) + bpr, nb)
-            if l > f:
-                stripes.append((2 * np.random.default_rng(100 + r)
-                                .random(((l - f) * w, k)) - 1)
-                               .astype(np.float32))
-        X_full = np.concatenate(stripes)
-        rows_l, cols_l, data_l = [], [], []
-        for c in range(nb):
-            ip, ix, dv = generate_block('row0', w, 10 * c + 1, gen_device, band)
-            rows_l.append(np.repeat(np.arange(w), np.diff(ip)))
-            cols_l.append(ix.astype(np.int64) + c * w)
+    # the deferred C_0 collective of the LAST step is waited only at the
+    # next step's consumer — run one more step so every rank's stripe
+    # (incl. the head refresh) reflects iteration 3... simpler: compare
+    # against FOUR golden iterations after a fourth step.
+    arrow.step()
+    eng0.set_features(eng0.result_tile())
+    iters = 4
+    # rank-LOCAL comparison (no collective in the check itself: gloo has
+    # no CUDA all_gather); every rank rebuilds the full A and X from the
+    # deterministic generators and checks its own stripe.
+    eng0._flush_x0_pipeline()  # drain the deferred collective + head copy
+    C_local = eng0.feature_tile().cpu().numpy()
+    first, last = eng0.first_block, eng0.last_block
+    gen_device = 'cuda' if use_gpu else 'cpu'
+    bpr = -(-nb // world)
+    stripes = []
+    for r in range(world):
+        f, l = min(r * bpr, nb), min(min(r * bpr, nb) + bpr, nb)
+        if l > f:
+            stripes.append((2 * np.random.default_rng(100 + r)
+                            .random(((l - f) * w, k)) - 1)
+                           .astype(np.float32))
+    X_full = np.concatenate(stripes)
+    rows_l, cols_l, data_l = [], [], []
+    for c in range(nb):
+        ip, ix, dv = generate_block('row0', w, 10 * c + 1, gen_device, band)
+        rows_l.append(np.repeat(np.arange(w), np.diff(ip)))
+        cols_l.append(ix.astype(np.int64) + c * w)
+        data_l.append(dv)
+    for r in range(1, nb):
+        for role, seed, coff in (('diag', 10 * r + 2, r * w),
+                                 ('col', 10 * r + 3, 0)):
+            ip, ix, dv = generate_block(role, w, seed, gen_device, band)
+            rows_l.append(np.repeat(np.arange(w), np.diff(ip)) + r * w)
+            cols_l.append(ix.astype(np.int64) + coff)
             data_l.append(dv)
-        for r in range(1, nb):
-            for role, seed, coff in (('diag', 10 * r + 2, r * w),
-                                     ('col', 10 * r + 3, 0)):
-                ip, ix, dv = generate_block(role, w, seed, gen_device, band)
-                rows_l.append(np.repeat(np.arange(w), np.diff(ip)) + r * w)
-                cols_l.append(ix.astype(np.int64) + coff)
-                data_l.append(dv)
-        A = sparse.csr_matrix(
-            (np.concatenate(data_l),
-             (np.concatenate(rows_l), np.concatenate(cols_l))),
-            shape=(nb * w, nb * w))
-        G = X_full
-        for _ in range(3):
-            G = A @ G
-        err = float(np.abs(C - G).max())
-        scale = max(1.0, float(np.abs(G).max()))
-        status = {"check": "ok" if err <= 1e-4 * scale else "FAIL",
-                  "max_abs_err": err, "scale": scale,
-                  "rel_err": err / scale, "world": world, "rows": nb * w,
-                  "k": k}
+    A = sparse.csr_matrix(
+        (np.concatenate(data_l),
+         (np.concatenate(rows_l), np.concatenate(cols_l))),
+        shape=(nb * w, nb * w))
+    G = X_full
+    for _ in range(iters):
+        G = A @ G
+    G_local = G[first * w:last * w]
+    n_cmp = min(C_local.shape[0], G_local.shape[0])
+    err = float(np.abs(C_local[:n_cmp] - G_local[:n_cmp]).max()) \
+        if n_cmp else 0.0
+    scale = max(1.0, float(np.abs(G).max()))
+    ok = err <= 1e-4 * scale
+    flag = torch.tensor([0.0 if ok else 1.0])
+    comm.allreduce_max_(flag)
+    status = {"check": "ok" if flag.item() == 0 else "FAIL",
+              "rank_max_abs_err": err, "scale": scale,
+              "rel_err": err / scale, "world": world, "rows": nb * w,
+              "k": k, "iters": iters}
+    if comm.rank == 0:
         print(json.dumps(status))
+    else:
+        print(f"# rank {comm.rank} check err={err:.3e}", file=sys.stderr)
     return status
 
 
@@ -554,12 +573,16 @@ def main():
         import gc
         graph = None
         X = None
-        eng0 = None
+        eng = None   # loop variable from the allreduce_x0 setup still
+        eng0 = None  # pins an engine (structures + 51 GB stripes)
         arrow = None
         one_step = None  # the closure holds arrow/eng0
         events.clear()
         gc.collect()
         torch.cuda.empty_cache()
+        free, total = torch.cuda.mem_get_info()
+        print(f"# traffic probe: freed workload, {free/2**30:.0f} GiB free",
+              file=sys.stderr)
         roofline['traffic'] = traffic_probe(args)
 
     if rank == 0:

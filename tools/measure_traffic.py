@@ -37,8 +37,17 @@ def run_pass(counters, outdir, bench_args):
            'spmm_kernel', '-d', outdir, '--',
            sys.executable, os.path.join(REPO, 'bench.py'), *bench_args,
            '--no-cpu-baseline', '--graph', 'off']
-    subprocess.run(cmd, check=True, cwd='/tmp', env=env,
-                   stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    os.makedirs(outdir, exist_ok=True)
+    log = os.path.join(outdir, 'pass.log')
+    try:
+        with open(log, 'w') as lf:
+            subprocess.run(cmd, check=True, cwd='/tmp', env=env,
+                           stdout=lf, stderr=lf)
+    except subprocess.CalledProcessError:
+        sys.stderr.write("---- rocprofv3 pass log tail ----\n")
+        with open(log) as lf:
+            sys.stderr.write(''.join(lf.readlines()[-25:]))
+        raise
     db = glob.glob(f'{outdir}/runc/*_results.db') + glob.glob(f'{outdir}/*_results.db')
     con = sqlite3.connect(db[0])
     cur = con.cursor()
