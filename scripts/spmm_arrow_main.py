@@ -12,30 +12,34 @@ from arrow_matrix_amd.common import utils
 
 
 def main() -> None:
-    parser = argparse.ArgumentParser(description='Benchmark the SpMM')
+    # Flags and defaults are the reference CLI's (spmm_arrow_main.py:10-31);
+    # help text reworded here.
+    parser = argparse.ArgumentParser(description='Iterated arrow-SpMM benchmark')
     parser.add_argument('-f', '--path', type=str, default=None,
-                        help='The filename prefix of the decomposed graph. '
-                             'If none, synthetic data is generated.')
+                        help='decomposed-graph file prefix; omit to run on '
+                             'generated synthetic data')
     parser.add_argument('-w', '--width', type=int, default=0,
-                        help='Width of the decomposition / Height of the blocks.')
+                        help='arrow width (block height) of the decomposition')
     parser.add_argument('-c', '--features', type=int, default=16,
-                        help='Width of the decomposition / Height of the blocks.')
+                        help='feature columns k of the dense operand X')
     parser.add_argument('-b', '--blocked', type=utils.str2bool, nargs="?", default=True,
-                        help='If true, the matrix has only one block diagonal,')
+                        help='block-diagonal decomposition (False: banded '
+                             'with +-1 off-diagonal blocks)')
     parser.add_argument('-i', '--device', type=str, default='gpu',
-                        help='Device to use for the MM. Either cpu or gpu.')
+                        help="compute device: 'gpu' (HIP kernels) or 'cpu' "
+                             "(scipy, the parity reference)")
     parser.add_argument('-z', '--iterations', type=int, default=1,
-                        help='Number of SpMM iteration to run.')
+                        help='number of X <- A @ X iterations')
     parser.add_argument('-r', '--ranksperside', type=int, default=3,
-                        help='Number of Ranks per Side (For synthetic data only)')
+                        help='synthetic data only: block-rows per matrix')
     parser.add_argument('-m', '--ba_neighbors', type=int, default=3,
-                        help='Number of neighbors per bertex (For synthetic data only)')
+                        help='synthetic data only: average neighbors per vertex')
     parser.add_argument('-s', '--slim', type=utils.str2bool, nargs="?", default=True,
-                        help='If true, the decomposition onto ranks is "slim" '
-                             'assigning one rank per row-block.')
+                        help='slim layout (one rank per block-row); False '
+                             'selects the banded ArrowMPI variant')
     parser.add_argument('-n', '--npy', type=utils.str2bool, nargs="?", default=True,
-                        help='If true, the decomposition is loaded from the '
-                             'indices / indptr files.')
+                        help='load the .npy (indptr/indices/data) format; '
+                             'False loads the legacy .npz format')
 
     args = vars(parser.parse_args())
     comm = default_comm()
