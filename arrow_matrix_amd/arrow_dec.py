@@ -198,11 +198,28 @@ class ArrowDecompositionMPI:
         # rows L=2: fold 3060 vs sequential 3155 GF/s, and the gap grows
         # with size — profiles/r02_ab2_sweep.log L2_fold/L2_seq).
         n0_rows = int(self.n_blocks[0]) * self.width
+        auto_fold = False
+        if (fold_env == 'auto' and self._fold_maps is not None
+                and self.comm.size == 1 and self.device == 'gpu'):
+            auto_fold = n0_rows <= 4_000_000
+            if not auto_fold:
+                # fold also when the SEQUENTIAL layout's extra per-part
+                # X/C buffers would not fit in HBM (fold keeps only part
+                # 0's buffers): e.g. L=2 at 100M rows x k=128 needs ~260
+                # GB sequentially but ~175 GB folded
+                try:
+                    import torch as _t
+                    free, _ = _t.cuda.mem_get_info()
+                    k = self._n_feature_columns
+                    extra = sum(
+                        4 * k * (2 * eng.n_owned * self.width + 2 * self.width)
+                        for eng in self.engines[1:])
+                    auto_fold = extra > 0.5 * free
+                except Exception:
+                    pass
         want_fold = (self._fold_maps is not None and self.comm.size == 1
                      and len(blocked) > 1
-                     and (fold_env == '1'
-                          or (fold_env == 'auto' and self.device == 'gpu'
-                              and n0_rows <= 4_000_000)))
+                     and (fold_env == '1' or auto_fold))
         if want_fold:
             self.engines[0].load_sparse_matrix_from_blocks(blocked[0])
             folded, dropped = self._build_folded(blocked)
