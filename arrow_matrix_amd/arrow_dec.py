@@ -209,12 +209,14 @@ class ArrowDecompositionMPI:
                 # GB sequentially but ~175 GB folded
                 try:
                     import torch as _t
-                    free, _ = _t.cuda.mem_get_info()
+                    free, total = _t.cuda.mem_get_info()
                     k = self._n_feature_columns
-                    extra = sum(
-                        4 * k * (2 * eng.n_owned * self.width + 2 * self.width)
-                        for eng in self.engines[1:])
-                    auto_fold = extra > 0.5 * free
+                    need = [4 * k * (2 * eng.n_owned * self.width
+                                     + 2 * self.width)
+                            for eng in self.engines]
+                    # sequential fits only if ALL parts' X/C buffers plus a
+                    # 15% allocator/workspace margin fit what is free now
+                    auto_fold = sum(need) + 0.15 * total > free
                 except Exception:
                     pass
         want_fold = (self._fold_maps is not None and self.comm.size == 1
