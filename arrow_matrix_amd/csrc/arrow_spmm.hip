@@ -40,7 +40,7 @@
 
 #include "../../include/arrow_spmm.h"
 
-#define ARROW_ABI_VERSION 1000
+#define ARROW_ABI_VERSION 1001  // round 2: set_qblocks, create_opts flags bit 2, wave-grab scheduler
 
 namespace {
 
