@@ -222,6 +222,11 @@ class ArrowDecompositionMPI:
         want_fold = (self._fold_maps is not None and self.comm.size == 1
                      and len(blocked) > 1
                      and (fold_env == '1' or auto_fold))
+        if os.environ.get('ARROW_FOLD_DEBUG') == '1':
+            import sys as _sys
+            print(f"# fold decision: want={want_fold} auto={auto_fold} "
+                  f"env={fold_env} maps={self._fold_maps is not None} "
+                  f"P={self.comm.size} L={len(blocked)}", file=_sys.stderr)
         if want_fold:
             self.engines[0].load_sparse_matrix_from_blocks(blocked[0])
             folded, dropped = self._build_folded(blocked)
