@@ -215,8 +215,9 @@ class ArrowDecompositionMPI:
                                      + 2 * self.width)
                             for eng in self.engines]
                     # sequential fits only if ALL parts' X/C buffers plus a
-                    # 15% allocator/workspace margin fit what is free now
-                    auto_fold = sum(need) + 0.15 * total > free
+                    # 25% margin (caller-side feature stripe, structures,
+                    # allocator slack) fit what is free now
+                    auto_fold = sum(need) + 0.25 * total > free
                 except Exception:
                     pass
         want_fold = (self._fold_maps is not None and self.comm.size == 1
