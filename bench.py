@@ -361,6 +361,38 @@ def parity_check(arrow, comm, w, nb, k, band, use_gpu, world):
     return status
 
 
+def measure_hbm_peak(reps=10, gib=2.0):
+    """Streaming HBM peak on THIS box (the roofline
+    denominator is the measured per-GPU peak, spec stated alongside).
+    Best of copy/scale/triad over fp32 buffers, HIP-event timed."""
+    import torch
+    n = int(gib * (1 << 30) / 4)
+    x = torch.rand(n, device='cuda')
+    z = torch.rand(n, device='cuda')
+    y = torch.empty(n, device='cuda')
+    for _ in range(3):
+        y.copy_(x)
+    torch.cuda.synchronize()
+    s, e = torch.cuda.Event(True), torch.cuda.Event(True)
+
+    def best(fn, bpe):
+        t = float('inf')
+        for _ in range(reps):
+            s.record()
+            fn()
+            e.record()
+            torch.cuda.synchronize()
+            t = min(t, s.elapsed_time(e))
+        return bpe * n / (t * 1e-3) / 1e9
+
+    peak = max(best(lambda: y.copy_(x), 8),
+               best(lambda: torch.mul(x, 1.0001, out=y), 8),
+               best(lambda: torch.add(x, z, alpha=1.0001, out=y), 12))
+    del x, z, y
+    torch.cuda.empty_cache()
+    return round(peak, 1)
+
+
 def main():
     args = build_parser().parse_args()
     import torch
@@ -397,6 +429,12 @@ def main():
     k = args.features
     L = args.parts
     gen_device = 'cuda' if use_gpu else 'cpu'
+
+    hbm_peak_meas = None
+    if use_gpu and rank == 0 and world <= 1 and not args.check:
+        hbm_peak_meas = measure_hbm_peak()
+        print(f"# measured streaming HBM peak: {hbm_peak_meas} GB/s "
+              f"(spec 8000)", file=sys.stderr)
 
     t0 = time.perf_counter()
     grids, first, last = build_blocks_for_rank(comm.rank, comm.size, w, nb, L,
@@ -561,6 +599,10 @@ def main():
             "launches": len(events),
             "avg_launch_ms": round(total_ms / len(events), 4),
         }
+        if hbm_peak_meas:
+            # §8d's denominator: the box's own measured streaming peak
+            roofline["peak_measured"] = hbm_peak_meas
+            roofline["frac_measured"] = round(achieved / hbm_peak_meas, 4)
 
     cpu_base = None
     if rank == 0 and world <= 1 and not args.no_cpu_baseline:
