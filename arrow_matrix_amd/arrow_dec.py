@@ -190,7 +190,13 @@ class ArrowDecompositionMPI:
         wrote)."""
         assert len(blocked) == self.decomposition_length
         import os
+        # ARROW_FOLD: '0' sequential exchange, '1' full fold, '2' row fold
+        # (backward cascade only), 'auto' (default): full fold in the
+        # launch-bound / memory-pressure regimes, row fold otherwise at
+        # world=1 on GPU (measured: row fold keeps part-i band locality
+        # and removes the backward passes)
         fold_env = os.environ.get('ARROW_FOLD', 'auto')
+        self._fold_cols = True
         # auto: fold only in the LAUNCH-BOUND small regime (part-0 rows <=
         # 4M). At larger scale the sequential path wins: materialising the
         # permuted X deduplicates the ~deg-many reads each part-i entry
@@ -220,19 +226,29 @@ class ArrowDecompositionMPI:
                     auto_fold = sum(need) + 0.25 * total > free
                 except Exception:
                     pass
-        want_fold = (self._fold_maps is not None and self.comm.size == 1
-                     and len(blocked) > 1
-                     and (fold_env == '1' or auto_fold))
+        foldable = (self._fold_maps is not None and self.comm.size == 1
+                    and len(blocked) > 1
+                    and not any(getattr(e, 'banded', False)
+                                for e in self.engines))
+        # row fold (backward cascade only) is the default remaining-GPU
+        # choice at world=1: keeps part-i X locality via the materialised
+        # forward exchange while removing the backward passes
+        want_row_fold = (foldable and fold_env == '2'
+                         or (foldable and fold_env == 'auto'
+                             and self.device == 'gpu' and not auto_fold))
+        want_fold = foldable and (fold_env == '1' or auto_fold)
         if os.environ.get('ARROW_FOLD_DEBUG') == '1':
             import sys as _sys
-            print(f"# fold decision: want={want_fold} auto={auto_fold} "
-                  f"env={fold_env} maps={self._fold_maps is not None} "
+            print(f"# fold decision: full={want_fold} row={want_row_fold} "
+                  f"auto={auto_fold} env={fold_env} "
+                  f"maps={self._fold_maps is not None} "
                   f"P={self.comm.size} L={len(blocked)}", file=_sys.stderr)
         if want_fold:
             self.engines[0].load_sparse_matrix_from_blocks(blocked[0])
-            folded, dropped = self._build_folded(blocked)
+            folded, dropped = self._build_folded(blocked, fold_cols=True)
             if folded is not None:
                 self._folded = folded
+                self._fold_cols = True
                 return
             import warnings
             warnings.warn(
@@ -242,18 +258,38 @@ class ArrowDecompositionMPI:
             for eng, blocks in zip(self.engines[1:], blocked[1:]):
                 eng.load_sparse_matrix_from_blocks(blocks)
             return
+        if want_row_fold:
+            self.engines[0].load_sparse_matrix_from_blocks(blocked[0])
+            folded, _ = self._build_folded(blocked, fold_cols=False)
+            self._folded = folded
+            self._fold_cols = False
+            return
         for eng, blocks in zip(self.engines, blocked):
             eng.load_sparse_matrix_from_blocks(blocks)
 
-    def _build_folded(self, blocked):
-        """Build one re-indexed structure per part i >= 1: entry
-        (r, c, v) of part i becomes (R_i[r], M_i[c], v) — a direct
-        contribution C_0[R_i[r]] += v * X_0[M_i[c]] — so part i runs as one
-        beta=1 SpMM against part 0's buffers with NO permutation exchange.
-        Per output row the accumulation order equals the sequential path's
-        (each part-i row is summed as its own work item, then added once).
-        Returns (list, 0) or (None, n_dropped) when entries fall outside
-        the composed maps."""
+    def _build_folded(self, blocked, fold_cols=True):
+        """Build one re-indexed structure per part i >= 1.
+
+        fold_cols=True (FULL fold): entry (r, c, v) of part i becomes
+        (R_i[r], M_i[c], v) — a direct contribution
+        C_0[R_i[r]] += v * X_0[M_i[c]] — so part i runs as one beta=1 SpMM
+        against part 0's buffers with NO permutation exchange at all.
+        Returns (None, n_dropped) when entries fall outside the composed
+        maps (the exchange would read rows no sender wrote).
+
+        fold_cols=False (ROW fold): only the BACKWARD cascade folds —
+        entry (r, c, v) becomes (R_i[r], c, v) reading part i's OWN
+        exchanged features X_i (the forward permutation still
+        materialises X_i, preserving part i's banded X locality) and
+        accumulating straight into part 0's C. This removes the backward
+        gather + scatter-add passes (3 stripe sweeps) at the cost of one
+        C read-modify-write inside the launch. Rows with R_i < 0 are rows
+        whose results the sequential cascade DISCARDS (never sent
+        backwards) — dropped here identically, not an error.
+
+        Per output row the accumulation order equals the sequential
+        path's (each part-i row is summed as its own work item, then
+        added once)."""
         w = self.width
         eng0 = self.engines[0]
         be = eng0.backend
@@ -262,6 +298,7 @@ class ArrowDecompositionMPI:
         folded = []
         for i in range(1, self.decomposition_length):
             M_i, R_i = self._fold_maps[i]
+            ni = int(self.n_blocks[i]) * w
             rows_l, cols_l, data_l = [], [], []
             for br, rowlist in enumerate(blocked[i]):
                 for bc, blk in enumerate(rowlist):
@@ -277,11 +314,17 @@ class ArrowDecompositionMPI:
             cols = np.concatenate(cols_l) if cols_l else np.empty(0, np.int64)
             data = np.concatenate(data_l) if data_l else np.empty(0, np.float32)
             nr = R_i[np.clip(rows, 0, None)]
-            nc = M_i[np.clip(cols, 0, None)]
-            keep = (nr >= 0) & (nc >= 0)
-            n_drop = int(rows.size - keep.sum())
-            if n_drop:
-                return None, n_drop
+            if fold_cols:
+                nc = M_i[np.clip(cols, 0, None)]
+                keep = (nr >= 0) & (nc >= 0)
+                n_drop = int(rows.size - keep.sum())
+                if n_drop:
+                    return None, n_drop
+            else:
+                nc = cols
+                keep = nr >= 0  # discarded-by-the-cascade rows drop
+            nr, nc, data = nr[keep], nc[keep], data[keep]
+            n_cols = n0 if fold_cols else ni
             # group by target row; explicit row_ids keep writes exclusive
             order = np.argsort(nr, kind='stable')
             nr, nc, data = nr[order], nc[order], data[order]
@@ -289,23 +332,24 @@ class ArrowDecompositionMPI:
             indptr = np.zeros(uniq.size + 1, dtype=np.int64)
             np.cumsum(counts, out=indptr[1:])
             if gpu:
-                h = be.upload_arrays((uniq.size, n0), indptr,
+                h = be.upload_arrays((uniq.size, n_cols), indptr,
                                      nc.astype(np.int32), data,
                                      row_ids=uniq.astype(np.int64))
-                h.x_rows = n0
+                h.x_rows = n_cols
                 folded.append(h)
             else:
                 from scipy import sparse
                 csr = sparse.csr_matrix(
                     (data.astype(be.np_dtype), nc.astype(np.int64), indptr),
-                    shape=(uniq.size, n0))
+                    shape=(uniq.size, n_cols))
                 folded.append((csr, uniq))
         return folded, 0
 
     def zero_rhs(self, width: int, n_features: int, dtype=np.float32) -> None:
         for i, eng in enumerate(self.engines):
-            if self._folded is not None and i > 0:
-                continue  # folded parts have no buffers of their own
+            if self._folded is not None and self._fold_cols and i > 0:
+                continue  # FULL-folded parts have no buffers of their own
+                # (row-folded parts keep X/C for the forward exchange)
             eng.zero_rhs(width, n_features, dtype=dtype)
 
     # -- iteration -----------------------------------------------------------
@@ -368,22 +412,29 @@ class ArrowDecompositionMPI:
 
     def _step_folded(self) -> None:
         """Folded iteration: part 0's arrow SpMM, then each folded part as
-        one beta=1 SpMM reading part 0's pre-step X and accumulating into
-        part 0's fresh C — results identical to the sequential
-        propagate/spmm/aggregate cascade (same per-row sums; fp32 add order
-        across parts preserved: part i is added after part i-1's own
-        contribution, as the backward cascade does)."""
+        one beta=1 SpMM accumulating into part 0's fresh C — results
+        identical to the sequential propagate/spmm/aggregate cascade (same
+        per-row sums). FULL fold reads part 0's pre-step X through the
+        composed maps (no exchange at all); ROW fold first materialises
+        each part's X via the forward exchange (band locality kept) and
+        folds only the backward cascade into the launch's output rows."""
         eng0 = self.engines[0]
+        if not self._fold_cols:
+            tic = time.perf_counter()
+            self._propagate_features()
+            wb_logging.log({"spmm_bcast_time": time.perf_counter() - tic})
         X_pre = eng0.feature_tile()
         tic = time.perf_counter()
         eng0.spmm()
-        for h in self._folded:
+        for i, h in enumerate(self._folded, start=1):
+            X_op = X_pre if self._fold_cols \
+                else self.engines[i].feature_tile()
             if eng0.backend.device == 'cuda':
-                eng0._timed(lambda h=h: eng0.backend.spmm_block(
-                    h, X_pre, eng0.C_i, 1), h.nnz, h.shape[0], h.x_rows)
+                eng0._timed(lambda h=h, X_op=X_op: eng0.backend.spmm_block(
+                    h, X_op, eng0.C_i, 1), h.nnz, h.shape[0], h.x_rows)
             else:
                 csr, rid = h
-                eng0.C_i.numpy()[rid] += csr @ X_pre.numpy()
+                eng0.C_i.numpy()[rid] += csr @ X_op.numpy()
         # C was mutated after the C_0 head was captured: any cached X_0
         # (the allreduce_x0 fast path) is stale now
         eng0._x0_valid = False

@@ -183,3 +183,14 @@ def test_engine_folded_engages_and_matches_sequential(monkeypatch):
     folded, seq = run(True), run(False)
     for F, S in zip(folded, seq):
         np.testing.assert_allclose(F, S, rtol=2e-6, atol=2e-6)
+
+
+def test_engine_row_folded_cpu(monkeypatch):
+    """ROW-fold mode (ARROW_FOLD=2): forward exchange materialised, only
+    the backward cascade folded into part-i launches — must match golden."""
+    monkeypatch.setenv('ARROW_FOLD', '2')
+    for n_blocks, seed in ([4, 2], 3), ([3, 3], 4), ([4, 3, 2], 5):
+        decomp = synth.synth_arrow_decomposition(5, n_blocks, avg_deg=5, seed=seed)
+        results, goldens = _run_engine(decomp, 5, n_blocks, 8, iters=3, seed=seed)
+        for C, G in zip(results, goldens):
+            np.testing.assert_allclose(C, G, rtol=1e-4, atol=1e-4)

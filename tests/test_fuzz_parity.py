@@ -111,3 +111,11 @@ def test_fuzz_case_folded(case, monkeypatch):
     (arrow_dec._build_folded) on CPU."""
     monkeypatch.setenv('ARROW_FOLD', '1')
     test_fuzz_case(case)
+
+
+@pytest.mark.parametrize("case", [c for c in CASES if len(c['n_blocks']) > 1])
+def test_fuzz_case_row_folded(case, monkeypatch):
+    """The L>1 sweep through the ROW-fold path (backward cascade folded,
+    forward exchange materialised) on CPU."""
+    monkeypatch.setenv('ARROW_FOLD', '2')
+    test_fuzz_case(case)

@@ -258,3 +258,20 @@ def test_engine_gpu_folded_matches_sequential(monkeypatch):
     folded, seq = run(True), run(False)
     for F, S in zip(folded, seq):
         np.testing.assert_allclose(F, S, rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_engine_gpu_row_folded_matches_golden(monkeypatch):
+    """ROW-fold (ARROW_FOLD=2) on GPU vs oracle golden at L=2/L=3."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    monkeypatch.setenv('ARROW_FOLD', '2')
+    from tests.test_engine_cpu import _run_engine
+    from arrow_matrix_amd import synth
+    for n_blocks, seed in ([4, 2], 2), ([4, 3, 2], 3):
+        decomp = synth.synth_arrow_decomposition(50, n_blocks, avg_deg=8,
+                                                 seed=seed)
+        results, goldens = _run_engine(decomp, 50, n_blocks, 8, iters=2,
+                                       device='gpu', seed=seed)
+        for C, G in zip(results, goldens):
+            np.testing.assert_allclose(C, G, rtol=1e-4, atol=1e-4)
