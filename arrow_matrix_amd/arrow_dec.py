@@ -323,13 +323,24 @@ class ArrowDecompositionMPI:
             else:
                 nc = cols
                 keep = nr >= 0  # discarded-by-the-cascade rows drop
+            rows_k = rows[keep]
             nr, nc, data = nr[keep], nc[keep], data[keep]
             n_cols = n0 if fold_cols else ni
-            # group by target row; explicit row_ids keep writes exclusive
-            order = np.argsort(nr, kind='stable')
+            # STRUCTURE ORDER decides the queue scheduler's X window. FULL
+            # fold reads part-0 X through the composed map (random either
+            # way) — group by TARGET row so C writes walk in order. ROW
+            # fold reads part i's own X — keep SOURCE-row order so the
+            # banded X window survives (the scatter lives in row_ids;
+            # target-ordered was measured -17% at 20M L=2); writes stay
+            # exclusive either way (R_i is injective).
+            key = nr if fold_cols else rows_k
+            order = np.argsort(key, kind='stable')
+            key_s = key[order]
             nr, nc, data = nr[order], nc[order], data[order]
-            uniq, counts = np.unique(nr, return_counts=True)
-            indptr = np.zeros(uniq.size + 1, dtype=np.int64)
+            uniq_key, starts, counts = np.unique(key_s, return_index=True,
+                                                 return_counts=True)
+            uniq = nr[starts]  # target row of each structure row
+            indptr = np.zeros(uniq_key.size + 1, dtype=np.int64)
             np.cumsum(counts, out=indptr[1:])
             if gpu:
                 h = be.upload_arrays((uniq.size, n_cols), indptr,
