@@ -429,18 +429,28 @@ class ArrowSlimMPI(ArrowMatrix):
             # busts 288 GB of HBM must fail with a sizing message, not a
             # mid-iteration OOM (VERDICT r1 "missing" #4; the reference's
             # own k-tiling, spmm_petsc.py:323-395, is moot at this
-            # capacity but the guard is not)
+            # capacity but the guard is not). Only bytes that would be
+            # NEWLY allocated count (a repeated zero_rhs with matching
+            # shapes reuses the buffers).
             import torch as _t
             w_, k_ = number_of_rows_per_rank, number_of_columns
-            need = 4 * k_ * (2 * max(self.n_owned, 1) * w_ + 2 * w_
-                             + (2 * w_ if self.banded else 0))
-            free, total = _t.cuda.mem_get_info()
-            if need > free:
-                raise MemoryError(
-                    f"zero_rhs: feature/result buffers need {need/2**30:.1f}"
-                    f" GiB but only {free/2**30:.1f} GiB of HBM are free "
-                    f"(k={k_}, width={w_}, {self.n_owned} owned blocks). "
-                    f"Shard over more GPUs or reduce k.")
+            stripe_sh = (max(self.n_owned, 1) * w_, k_)
+            shapes = [('C_i', stripe_sh), ('X_i', stripe_sh),
+                      ('C_0', (w_, k_)), ('X_0', (w_, k_))]
+            if self.banded:
+                shapes += [('X_halo_lo', (w_, k_)), ('X_halo_hi', (w_, k_))]
+            need = sum(4 * sh[0] * sh[1] for name, sh in shapes
+                       if getattr(self, name) is None
+                       or tuple(getattr(self, name).shape) != sh)
+            if need:
+                free, total = _t.cuda.mem_get_info()
+                if need > free:
+                    raise MemoryError(
+                        f"zero_rhs: feature/result buffers need "
+                        f"{need/2**30:.1f} GiB but only {free/2**30:.1f} "
+                        f"GiB of HBM are free (k={k_}, width={w_}, "
+                        f"{self.n_owned} owned blocks). Shard over more "
+                        f"GPUs or reduce k.")
         self.width = number_of_rows_per_rank
         w, k = number_of_rows_per_rank, number_of_columns
         stripe = (max(self.n_owned, 1) * w, k)
