@@ -143,3 +143,62 @@ def test_golden_fixtures_are_fresh(tmp_path, monkeypatch):
     assert sorted(fresh.files) == sorted(committed.files)
     for key in committed.files:
         np.testing.assert_array_equal(fresh[key], committed[key], err_msg=key)
+
+
+def test_fold_maps_match_exchange_semantics():
+    """The composed fold maps (arrow_dec._compute_fold_maps) must agree
+    with what the forward/backward exchanges actually do at P=1: routing
+    X through the exchange chain equals gathering through M_i, and the
+    backward cascade's accumulation equals scattering through R_i."""
+    import numpy as np
+    from arrow_matrix_amd import synth, tables
+    from arrow_matrix_amd.arrow_dec import ArrowDecompositionMPI
+
+    rng = np.random.default_rng(77)
+    for n_blocks, width, seed in ([4, 2], 5, 1), ([3, 3, 2], 4, 2):
+        decomp = synth.synth_arrow_decomposition(width, n_blocks, avg_deg=4,
+                                                 seed=seed)
+        perms = [np.asarray(p) for _, p in decomp]
+        nb = np.asarray(n_blocks)
+        _, to_prev, to_next = tables.pad_and_compose_permutations(
+            perms, nb, width)
+        # single process: each part's slice is its first nb_i*width rows
+        # (the loader's slicing, arrow_dec.py::load_decomposition_new)
+        tp_s = [None if t is None else t[:n_blocks[i] * width]
+                for i, t in enumerate(to_prev)]
+        tn_s = [None if t is None else t[:n_blocks[i] * width]
+                for i, t in enumerate(to_next)]
+        arrow = ArrowDecompositionMPI.initialize(
+            None, nb, tp_s, tn_s, width, 3, device='cpu')
+        assert arrow._fold_maps is not None
+        n0 = n_blocks[0] * width
+        X0 = rng.random((n0, 3)).astype(np.float32)
+        # forward chain: X_i via successive routing == gather through M_i
+        X_prev = X0
+        for i in range(1, len(n_blocks)):
+            ni = n_blocks[i] * width
+            tn = tn_s[i - 1]
+            X_i = np.zeros((ni, 3), np.float32)
+            valid = tn < ni
+            X_i[tn[valid]] = X_prev[valid]
+            M_i, R_i = arrow._fold_maps[i]
+            X_via_map = np.zeros((ni, 3), np.float32)
+            m_ok = M_i >= 0
+            X_via_map[m_ok] = X0[M_i[m_ok]]
+            np.testing.assert_array_equal(X_i, X_via_map)
+            # backward single hop vs R map through the chain: scatter C_i
+            # through the cascade == scatter through R_i into part 0
+            C_i = rng.random((ni, 3)).astype(np.float32)
+            acc = C_i
+            for j in range(i, 0, -1):
+                tp = tp_s[j]
+                nj = n_blocks[j - 1] * width
+                nxt = np.zeros((nj, 3), np.float32)
+                v = tp[:acc.shape[0]] < nj
+                nxt[tp[:acc.shape[0]][v]] = acc[v]
+                acc = nxt
+            via_R = np.zeros((n0, 3), np.float32)
+            r_ok = R_i >= 0
+            via_R[R_i[r_ok]] = C_i[r_ok]
+            np.testing.assert_array_equal(acc, via_R)
+            X_prev = X_i
